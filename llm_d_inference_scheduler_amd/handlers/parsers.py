@@ -1,0 +1,208 @@
+"""Request/response parsers (L2)
+(parity: pkg/epp/framework/plugins/requesthandling/parsers/* and the parser
+mux pkg/epp/handlers/parsers.go:28; unified body type
+requesthandling/types.go:65).
+
+* openai-parser: /v1/completions, /v1/chat/completions, /v1/responses,
+  /v1/conversations, /v1/embeddings JSON -> LLMRequest; SSE + JSON usage
+  parse on the response side (parsers/openai/openai.go:34-139).
+* passthrough-parser: raw bytes, Skip -> fallback random-endpoint routing.
+* vertexai-parser: Google Vertex AI payload shape.
+* vllm-grpc-parser: N/A on this node — the reference proxies an external
+  vLLM gRPC engine; here the engines are in-process (docs/PARITY.md).
+"""
+import json
+import uuid
+from dataclasses import dataclass, field
+from typing import Any, Dict, List, Optional
+
+from ..scheduling.types import LLMRequest, MultiModalItem
+from ..plugins.registry import register_plugin
+from ..plugins.interface import Plugin
+
+FAIRNESS_ID_HEADER = "x-gateway-inference-fairness-id"
+OBJECTIVE_HEADER = "x-gateway-inference-objective"
+MODEL_REWRITE_HEADER = "x-gateway-model-name-rewrite"
+
+
+@dataclass
+class Usage:
+    prompt_tokens: int = 0
+    completion_tokens: int = 0
+    cached_tokens: int = 0
+    ttft_ms: Optional[float] = None
+    tpot_ms: Optional[float] = None
+    e2e_ms: Optional[float] = None
+
+    def to_openai(self) -> Dict[str, Any]:
+        return {
+            "prompt_tokens": self.prompt_tokens,
+            "completion_tokens": self.completion_tokens,
+            "total_tokens": self.prompt_tokens + self.completion_tokens,
+            "prompt_tokens_details": {"cached_tokens": self.cached_tokens},
+        }
+
+
+@dataclass
+class ParseResult:
+    request: Optional[LLMRequest] = None
+    skip: bool = False          # Skip -> random-endpoint fallback routing
+    error: Optional[str] = None
+
+
+class Parser(Plugin):
+    def parse_request(self, body: bytes, headers: Dict[str, str],
+                      path: str = "") -> ParseResult:
+        raise NotImplementedError
+
+    def parse_response_usage(self, body: bytes,
+                             streaming: bool) -> Optional[Usage]:
+        return None
+
+
+def _headers_into(req: LLMRequest, headers: Dict[str, str]) -> None:
+    req.headers.update(headers)
+    req.fairness_id = headers.get(FAIRNESS_ID_HEADER, "")
+    req.objective_name = headers.get(OBJECTIVE_HEADER, "")
+
+
+def _extract_mm_items(messages: List[Dict[str, Any]]) -> List[MultiModalItem]:
+    items: List[MultiModalItem] = []
+    for msg in messages:
+        content = msg.get("content")
+        if not isinstance(content, list):
+            continue
+        for part in content:
+            kind = part.get("type", "")
+            if kind == "image_url":
+                url = part.get("image_url", {})
+                items.append(MultiModalItem("image_url",
+                                            url.get("url", "") if isinstance(url, dict) else str(url)))
+            elif kind == "video_url":
+                items.append(MultiModalItem("video_url",
+                                            part.get("video_url", {}).get("url", "")))
+            elif kind == "input_audio":
+                items.append(MultiModalItem("input_audio",
+                                            part.get("input_audio", {}).get("data", "")[:64]))
+    return items
+
+
+@register_plugin("openai-parser")
+class OpenAIParser(Parser):
+    def parse_request(self, body: bytes, headers: Dict[str, str],
+                      path: str = "") -> ParseResult:
+        try:
+            data = json.loads(body)
+        except (ValueError, TypeError) as e:
+            return ParseResult(error=f"invalid JSON: {e}")
+        if not isinstance(data, dict) or "model" not in data:
+            return ParseResult(error="missing model")
+        req = LLMRequest(
+            request_id=headers.get("x-request-id", uuid.uuid4().hex),
+            model=str(data["model"]),
+            raw_body=data,
+        )
+        if "/embeddings" in path or "input" in data and "messages" not in data \
+                and "prompt" not in data and path.endswith("embeddings"):
+            req.is_embedding = True
+        if "prompt" in data:
+            p = data["prompt"]
+            req.prompt = p if isinstance(p, str) else " ".join(map(str, p))
+        if "messages" in data and isinstance(data["messages"], list):
+            req.messages = data["messages"]
+            req.mm_items = _extract_mm_items(req.messages)
+        if "input" in data and not req.prompt and not req.messages:
+            inp = data["input"]
+            req.prompt = inp if isinstance(inp, str) else " ".join(map(str, inp))
+            req.is_embedding = "/embeddings" in path
+        req.max_tokens = int(data.get("max_tokens",
+                                      data.get("max_completion_tokens", 16)))
+        req.temperature = float(data.get("temperature", 0.0))
+        req.streaming = bool(data.get("stream", False))
+        req.session_id = str(data.get("user", "") or "")
+        _headers_into(req, headers)
+        return ParseResult(request=req)
+
+    def parse_response_usage(self, body: bytes,
+                             streaming: bool) -> Optional[Usage]:
+        """JSON body or SSE stream -> usage (response.go:45-63)."""
+        try:
+            if streaming:
+                usage = None
+                for line in body.split(b"\n"):
+                    line = line.strip()
+                    if not line.startswith(b"data:"):
+                        continue
+                    payload = line[5:].strip()
+                    if payload == b"[DONE]":
+                        break
+                    d = json.loads(payload)
+                    if d.get("usage"):
+                        usage = d["usage"]
+                d = {"usage": usage} if usage else {}
+            else:
+                d = json.loads(body)
+            u = d.get("usage") or {}
+            if not u:
+                return None
+            details = u.get("prompt_tokens_details") or {}
+            return Usage(prompt_tokens=int(u.get("prompt_tokens", 0)),
+                         completion_tokens=int(u.get("completion_tokens", 0)),
+                         cached_tokens=int(details.get("cached_tokens", 0)))
+        except (ValueError, TypeError):
+            return None
+
+
+@register_plugin("passthrough-parser")
+class PassthroughParser(Parser):
+    """Raw bytes, skip parsing -> random-endpoint fallback (parsers/passthrough)."""
+
+    def parse_request(self, body, headers, path=""):
+        return ParseResult(skip=True)
+
+
+@register_plugin("vertexai-parser")
+class VertexAIParser(Parser):
+    """Google Vertex AI payloads (parsers/vertexai): instances/contents shape."""
+
+    def parse_request(self, body: bytes, headers: Dict[str, str],
+                      path: str = "") -> ParseResult:
+        try:
+            data = json.loads(body)
+        except (ValueError, TypeError) as e:
+            return ParseResult(error=f"invalid JSON: {e}")
+        model = str(data.get("model", "")) or headers.get("x-vertex-model", "")
+        if not model:
+            return ParseResult(error="missing model")
+        req = LLMRequest(
+            request_id=headers.get("x-request-id", uuid.uuid4().hex),
+            model=model, raw_body=data)
+        contents = data.get("contents") or []
+        parts: List[str] = []
+        for c in contents if isinstance(contents, list) else []:
+            for part in c.get("parts", []):
+                if "text" in part:
+                    parts.append(str(part["text"]))
+        req.prompt = "\n".join(parts)
+        gen_cfg = data.get("generationConfig") or {}
+        req.max_tokens = int(gen_cfg.get("maxOutputTokens", 16))
+        req.temperature = float(gen_cfg.get("temperature", 0.0))
+        _headers_into(req, headers)
+        return ParseResult(request=req)
+
+
+class ParserMux:
+    """Per content-type / path dispatch (handlers/parsers.go:28)."""
+
+    def __init__(self, default: Optional[Parser] = None):
+        self.default = default or OpenAIParser()
+        self.by_content_type: Dict[str, Parser] = {}
+
+    def register(self, content_type: str, parser: Parser) -> None:
+        self.by_content_type[content_type] = parser
+
+    def parse_request(self, body: bytes, headers: Dict[str, str],
+                      path: str = "") -> ParseResult:
+        ct = headers.get("content-type", "application/json").split(";")[0]
+        parser = self.by_content_type.get(ct, self.default)
+        return parser.parse_request(body, headers, path)

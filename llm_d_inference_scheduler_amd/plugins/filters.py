@@ -1,0 +1,130 @@
+"""Filter plugins (parity: pkg/epp/framework/plugins/scheduling/filter/*).
+
+Role filters key on the `llm-d.ai/role` label exactly like
+filter/bylabel/roles.go:9-48; on this node the label lives on the GPU-role
+endpoint instead of a pod.
+"""
+from typing import Dict, List
+
+from ..datalayer.endpoint import CONTEXT_LENGTH_RANGE_LABEL, Endpoint, Role
+from ..datalayer.attributes import LATENCY_PREDICTION_INFO, PREFIX_CACHE_MATCH_INFO
+from ..scheduling.types import SchedulingContext
+from .interface import Filter
+from .registry import register_plugin
+
+
+class _RoleFilter(Filter):
+    role: Role = Role.DECODE
+
+    def filter(self, ctx, endpoints: List[Endpoint]) -> List[Endpoint]:
+        return [ep for ep in endpoints if ep.has_role(self.role)]
+
+
+@register_plugin("decode-filter", aliases=["decode_filter"])
+class DecodeFilter(_RoleFilter):
+    role = Role.DECODE
+
+
+@register_plugin("prefill-filter", aliases=["prefill_filter"])
+class PrefillFilter(_RoleFilter):
+    role = Role.PREFILL
+
+
+@register_plugin("encode-filter", aliases=["encode_filter"])
+class EncodeFilter(_RoleFilter):
+    role = Role.ENCODE
+
+
+@register_plugin("label-selector-filter", aliases=["by-label-selector"])
+class LabelSelectorFilter(Filter):
+    """K8s-style label selector: parameters.selector = {key: value} with
+    set-style values "a|b" allowed (filter/bylabel)."""
+
+    def __init__(self, name: str = "", **params):
+        super().__init__(name, **params)
+        self.selector: Dict[str, str] = params.get("selector", {})
+
+    def filter(self, ctx, endpoints: List[Endpoint]) -> List[Endpoint]:
+        out = []
+        for ep in endpoints:
+            ok = True
+            for k, v in self.selector.items():
+                allowed = str(v).split("|")
+                if ep.metadata.labels.get(k) not in allowed:
+                    ok = False
+                    break
+            if ok:
+                out.append(ep)
+        return out
+
+
+@register_plugin("prefix-cache-affinity-filter")
+class PrefixCacheAffinityFilter(Filter):
+    """Sticky-set pre-filter (filter/prefixcacheaffinity): when some
+    endpoints have a strong prefix match, keep only those; back off (keep
+    all) when the predicted TTFT of the sticky set exceeds the SLO gate."""
+
+    def __init__(self, name: str = "", **params):
+        super().__init__(name, **params)
+        self.min_ratio = float(params.get("minMatchRatio", 0.5))
+        self.ttft_backoff_ms = params.get("ttftBackoffMs")
+
+    def filter(self, ctx: SchedulingContext,
+               endpoints: List[Endpoint]) -> List[Endpoint]:
+        info = ctx.attributes.get(PREFIX_CACHE_MATCH_INFO)
+        if info is None or info.total_blocks == 0:
+            return endpoints
+        sticky = [ep for ep in endpoints
+                  if info.ratio(ep.name) >= self.min_ratio]
+        if not sticky:
+            return endpoints
+        if self.ttft_backoff_ms is not None:
+            lat = ctx.attributes.get(LATENCY_PREDICTION_INFO)
+            if lat is not None:
+                preds = [lat.predicted_ttft_ms.get(ep.name) for ep in sticky]
+                preds = [p for p in preds if p is not None]
+                if preds and min(preds) > float(self.ttft_backoff_ms):
+                    return endpoints  # TTFT back-off gate: fail open
+        return sticky
+
+
+@register_plugin("slo-headroom-tier-filter")
+class SLOHeadroomTierFilter(Filter):
+    """Positive/negative headroom tiers with epsilon exploration
+    (filter/sloheadroomtier): prefer endpoints whose predicted latency
+    leaves positive SLO headroom; with probability epsilon keep all."""
+
+    def __init__(self, name: str = "", **params):
+        super().__init__(name, **params)
+        self.epsilon = float(params.get("epsilon", 0.01))
+        import random
+        self._rng = random.Random(params.get("seed", 0xD15A))
+
+    def filter(self, ctx: SchedulingContext,
+               endpoints: List[Endpoint]) -> List[Endpoint]:
+        lat = ctx.attributes.get(LATENCY_PREDICTION_INFO)
+        if lat is None or not lat.ttft_headroom_ms:
+            return endpoints
+        if self._rng.random() < self.epsilon:
+            return endpoints  # exploration
+        positive = [ep for ep in endpoints
+                    if lat.ttft_headroom_ms.get(ep.name, -1.0) > 0
+                    and lat.tpot_headroom_ms.get(ep.name, 0.0) >= 0]
+        return positive or endpoints  # fail open to the negative tier
+
+
+@register_plugin("header-based-testing-filter")
+class HeaderBasedTestingFilter(Filter):
+    """Conformance/test-only filter (test/filter/): keep the endpoint named
+    by the `test-epp-endpoint-selection` request header."""
+
+    HEADER = "test-epp-endpoint-selection"
+
+    def filter(self, ctx: SchedulingContext,
+               endpoints: List[Endpoint]) -> List[Endpoint]:
+        want = ctx.request.headers.get(self.HEADER)
+        if not want:
+            return endpoints
+        names = set(want.split(","))
+        return [ep for ep in endpoints
+                if ep.name in names or ep.metadata.address in names]

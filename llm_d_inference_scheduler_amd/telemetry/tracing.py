@@ -1,0 +1,102 @@
+"""Lightweight tracing (parity: pkg/telemetry/tracing.go + pkg/common/observability/tracing).
+
+The reference initializes an OTel OTLP exporter and wraps the stream
+(`gateway.request`, server.go:179), director orchestration, profile-handler
+decisions and sidecar stages (incl. true_ttft_ms) in spans. There is no
+network egress here, so spans are recorded in-process (ring buffer +
+optional JSONL export) with the same span/attribute names; an OTLP exporter
+can be attached later without touching call sites.
+"""
+import json
+import threading
+import time
+from collections import deque
+from contextlib import contextmanager
+from dataclasses import dataclass, field
+from typing import Any, Dict, List, Optional
+
+
+@dataclass
+class Span:
+    name: str
+    start_ns: int
+    end_ns: int = 0
+    attributes: Dict[str, Any] = field(default_factory=dict)
+    parent: Optional[str] = None
+    trace_id: str = ""
+
+    def set_attribute(self, key: str, value: Any) -> None:
+        self.attributes[key] = value
+
+    @property
+    def duration_ms(self) -> float:
+        return (self.end_ns - self.start_ns) / 1e6
+
+
+class Tracer:
+    def __init__(self, service_name: str, capacity: int = 4096):
+        self.service_name = service_name
+        self._spans: deque = deque(maxlen=capacity)
+        self._lock = threading.Lock()
+        self._local = threading.local()
+        self.enabled = True
+
+    @contextmanager
+    def span(self, name: str, **attrs):
+        if not self.enabled:
+            yield _NOOP_SPAN
+            return
+        parent = getattr(self._local, "current", None)
+        s = Span(name=name, start_ns=time.monotonic_ns(),
+                 attributes=dict(attrs),
+                 parent=parent.name if parent else None)
+        prev = parent
+        self._local.current = s
+        try:
+            yield s
+        finally:
+            s.end_ns = time.monotonic_ns()
+            self._local.current = prev
+            with self._lock:
+                self._spans.append(s)
+
+    def finished_spans(self, name: Optional[str] = None) -> List[Span]:
+        with self._lock:
+            spans = list(self._spans)
+        if name is not None:
+            spans = [s for s in spans if s.name == name]
+        return spans
+
+    def export_jsonl(self, path: str) -> None:
+        with self._lock:
+            spans = list(self._spans)
+        with open(path, "w") as f:
+            for s in spans:
+                f.write(json.dumps({
+                    "name": s.name, "start_ns": s.start_ns, "end_ns": s.end_ns,
+                    "duration_ms": s.duration_ms, "parent": s.parent,
+                    "attributes": s.attributes}) + "\n")
+
+
+class _NoopSpan:
+    def set_attribute(self, key, value):
+        pass
+
+
+_NOOP_SPAN = _NoopSpan()
+_tracer: Optional[Tracer] = None
+
+
+def init_tracing(service_name: str = "llm-d-inference-scheduler-amd",
+                 enabled: bool = True) -> Tracer:
+    global _tracer
+    _tracer = Tracer(service_name)
+    _tracer.enabled = enabled
+    return _tracer
+
+
+def get_tracer() -> Tracer:
+    global _tracer
+    if _tracer is None:
+        _tracer = Tracer("llm-d-inference-scheduler-amd")
+    return _tracer

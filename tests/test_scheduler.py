@@ -1,0 +1,169 @@
+"""Scheduler core: profile run, native/python scorer parity, profile handlers."""
+import numpy as np
+import pytest
+
+from llm_d_inference_scheduler_amd.datalayer.attributes import (
+    PREFIX_CACHE_MATCH_INFO, PrefixCacheMatchInfo)
+from llm_d_inference_scheduler_amd.datalayer.datastore import make_endpoint
+from llm_d_inference_scheduler_amd.datalayer.endpoint import Metrics
+from llm_d_inference_scheduler_amd.plugins import register_all_plugins
+from llm_d_inference_scheduler_amd.plugins.registry import global_registry
+from llm_d_inference_scheduler_amd.plugins.profile_handlers import (
+    DisaggProfileHandler, PrefixBasedPDDecider, SingleProfileHandler)
+from llm_d_inference_scheduler_amd.scheduling.scheduler import (
+    Scheduler, SchedulerConfig, SchedulerProfile)
+from llm_d_inference_scheduler_amd.scheduling.types import (
+    LLMRequest, SchedulingContext)
+
+register_all_plugins()
+
+
+def mk(type_name, **params):
+    return global_registry.instantiate(type_name, **params)
+
+
+def set_metrics(ep, queue=0, running=0, kv=0.0):
+    m = Metrics(waiting_queue_size=queue, running_requests_size=running,
+                kv_cache_usage=kv)
+    ep.update_metrics(m)
+
+
+@pytest.fixture
+def ctx(request_factory):
+    return SchedulingContext(request=request_factory())
+
+
+class TestProfileRun:
+    def test_native_pick_least_loaded(self, endpoints, ctx):
+        for i, ep in enumerate(endpoints):
+            set_metrics(ep, queue=i * 2, kv=i * 0.2)
+        prof = SchedulerProfile(
+            name="p", scorers=[(mk("queue-scorer"), 1.0),
+                               (mk("kv-cache-utilization-scorer"), 1.0)],
+            picker=mk("max-score-picker"))
+        res = prof.run(ctx, endpoints)
+        assert res.target.name == "gpu0"
+        assert res.scores["gpu0"] == pytest.approx(2.0)
+
+    def test_python_native_parity(self, endpoints, ctx):
+        """Python scorer formulas == native formulas for the same state."""
+        for i, ep in enumerate(endpoints):
+            set_metrics(ep, queue=[3, 0, 7, 2][i], running=[1, 5, 2, 0][i],
+                        kv=[0.1, 0.9, 0.4, 0.0][i])
+        ctx.attributes[PREFIX_CACHE_MATCH_INFO] = PrefixCacheMatchInfo(
+            match_blocks={"gpu0": 2, "gpu2": 6}, total_blocks=8)
+        scorers = ["queue-scorer", "kv-cache-utilization-scorer",
+                   "prefix-cache-scorer", "running-requests-size-scorer",
+                   "load-aware-scorer"]
+        weights = [1.0, 2.0, 3.0, 0.5, 1.5]
+        plugins = [mk(s) for s in scorers]
+        # native run
+        prof = SchedulerProfile(name="n",
+                                scorers=list(zip(plugins, weights)),
+                                picker=mk("max-score-picker"))
+        res_native = prof.run(ctx, endpoints)
+        # python-side expected
+        expected = {ep.name: 0.0 for ep in endpoints}
+        for plugin, w in zip(plugins, weights):
+            smap = plugin.score(ctx, endpoints)
+            for name, v in smap.items():
+                expected[name] += w * min(1.0, max(0.0, v))
+        for name in expected:
+            assert res_native.scores[name] == pytest.approx(expected[name],
+                                                            abs=1e-5)
+
+    def test_filter_chain(self, endpoints, ctx):
+        prof = SchedulerProfile(name="p", filters=[mk("prefill-filter")],
+                                scorers=[(mk("queue-scorer"), 1.0)],
+                                picker=mk("max-score-picker"))
+        res = prof.run(ctx, endpoints)
+        assert res.target.name == "gpu3"  # only prefill-decode endpoint
+
+    def test_empty_filter_result(self, endpoints, ctx):
+        prof = SchedulerProfile(name="p", filters=[mk("encode-filter")],
+                                picker=mk("max-score-picker"))
+        res = prof.run(ctx, endpoints)
+        assert res.target is None
+
+    def test_max_endpoints(self, endpoints, ctx):
+        prof = SchedulerProfile(name="p",
+                                scorers=[(mk("queue-scorer"), 1.0)],
+                                picker=mk("max-score-picker"),
+                                max_endpoints=3)
+        res = prof.run(ctx, endpoints)
+        assert len(res.picks) == 3
+
+    def test_python_scorer_in_profile(self, endpoints, ctx):
+        """A non-native scorer flows through the `extra` array."""
+        sess = mk("session-affinity-scorer")
+        sess.remember("alice", "gpu2")
+        ctx.request.session_id = "alice"
+        prof = SchedulerProfile(name="p", scorers=[(sess, 5.0)],
+                                picker=mk("max-score-picker"))
+        res = prof.run(ctx, endpoints)
+        assert res.target.name == "gpu2"
+        assert res.scores["gpu2"] == pytest.approx(5.0)
+
+
+class TestSchedulerLoop:
+    def test_single_profile(self, endpoints, ctx):
+        prof = SchedulerProfile(name="default",
+                                scorers=[(mk("queue-scorer"), 1.0)],
+                                picker=mk("max-score-picker"))
+        sched = Scheduler(SchedulerConfig(
+            profiles={"default": prof},
+            profile_handler=SingleProfileHandler()))
+        res = sched.schedule(ctx, endpoints)
+        assert res.primary_profile == "default"
+        assert res.primary.target is not None
+
+    def _disagg_scheduler(self, non_cached_tokens=8):
+        decode = SchedulerProfile(name="decode", filters=[mk("decode-filter")],
+                                  scorers=[(mk("queue-scorer"), 1.0),
+                                           (mk("prefix-cache-scorer"), 2.0)],
+                                  picker=mk("max-score-picker"))
+        prefill = SchedulerProfile(name="prefill",
+                                   filters=[mk("prefill-filter")],
+                                   scorers=[(mk("queue-scorer"), 1.0)],
+                                   picker=mk("max-score-picker"))
+        handler = DisaggProfileHandler(
+            pdDecider=PrefixBasedPDDecider(nonCachedTokens=non_cached_tokens))
+        return Scheduler(SchedulerConfig(
+            profiles={"decode": decode, "prefill": prefill},
+            profile_handler=handler))
+
+    def test_disagg_fires_on_long_uncached_prompt(self, endpoints,
+                                                  request_factory):
+        req = request_factory()
+        req.prompt_tokens = list(range(1000))
+        ctx = SchedulingContext(request=req)
+        sched = self._disagg_scheduler(non_cached_tokens=512)
+        res = sched.schedule(ctx, endpoints)
+        assert res.primary_profile == "decode"
+        assert "prefill" in res.profile_results
+        assert res.profile_results["prefill"].target.name == "gpu3"
+        # PreRequest publishes the prefiller header
+        handler = sched.config.profile_handler
+        handler.pre_request(ctx, res, res.primary.target)
+        assert "x-prefiller-host-port" in req.headers
+
+    def test_disagg_skipped_on_short_prompt(self, endpoints, request_factory):
+        req = request_factory()
+        req.prompt_tokens = list(range(100))
+        ctx = SchedulingContext(request=req)
+        sched = self._disagg_scheduler(non_cached_tokens=512)
+        res = sched.schedule(ctx, endpoints)
+        assert "prefill" not in res.profile_results
+
+    def test_disagg_skipped_when_cached(self, endpoints, request_factory):
+        req = request_factory()
+        req.prompt_tokens = list(range(1000))
+        ctx = SchedulingContext(request=req)
+        # decode target has ~all blocks cached
+        ctx.attributes[PREFIX_CACHE_MATCH_INFO] = PrefixCacheMatchInfo(
+            match_blocks={ep: 62 for ep in
+                          ("gpu0", "gpu1", "gpu2", "gpu3")},
+            total_blocks=62, block_size_tokens=16)
+        sched = self._disagg_scheduler(non_cached_tokens=512)
+        res = sched.schedule(ctx, endpoints)
+        assert "prefill" not in res.profile_results

@@ -1,0 +1,401 @@
+"""NodeRunner — the per-rank driver wiring router + engines + xGMI transfer.
+
+Plays two reference roles at once:
+  * rank 0 is the EPP (runner.go:164 setup: datastore, datalayer, director,
+    scheduler, flow control) for the whole node;
+  * every rank is a worker "pod" plus its share of the pd-sidecar's stage
+    choreography (proxy.go / connector_nixlv2.go): prefill(max_tokens=1) ->
+    KV-handle return -> decode-with-handles, except the handles are pool
+    block indices and the KV moves over xGMI (parallel/transfer.py).
+
+All ranks run `step()` in lockstep; one control-plane mailbox exchange per
+step carries assignments, metrics, KV-ready notices and completions.
+"""
+import queue as queue_mod
+import time
+from concurrent.futures import ThreadPoolExecutor
+from dataclasses import dataclass, field
+from typing import Any, Dict, List, Optional
+
+import torch
+
+from ..config import LoadedConfig, load_config
+from ..datalayer.datastore import Datastore, make_endpoint
+from ..datalayer.endpoint import Metrics, Role
+from ..engine.worker import EngineRequest, EngineWorker, RequestOutput
+from ..flowcontrol import (BandConfig, FlowController, FlowRegistry,
+                           UtilizationSaturationDetector)
+from ..handlers.parsers import Usage
+from ..metrics import prom
+from ..models.configs import ModelConfig
+from ..parallel.mailbox import Mailbox
+from ..parallel.topology import NodeTopology
+from ..parallel.transfer import KVTransferEngine
+from ..plugins.producers import ApproxPrefixCacheProducer
+from ..requestcontrol import (AdmissionDenied, Director, EndpointCandidates,
+                              FlowControlAdmissionController,
+                              LegacyAdmissionController, RoutingDecision)
+from ..scheduling.scheduler import Scheduler
+from ..scheduling.types import LLMRequest
+from ..utils.logging import get_logger
+
+log = get_logger("node.runner")
+
+DEFAULT_EPP_CONFIG = """
+plugins:
+  - type: decode-filter
+  - type: prefill-filter
+  - type: queue-scorer
+  - type: kv-cache-utilization-scorer
+  - type: prefix-cache-scorer
+  - type: inflight-load-producer
+  - type: max-score-picker
+  - type: prefix-based-pd-decider
+    parameters: {nonCachedTokens: 512}
+  - type: disagg-profile-handler
+    parameters:
+      pdDecider: prefix-based-pd-decider
+schedulingProfiles:
+  - name: decode
+    plugins:
+      - {pluginRef: decode-filter}
+      - {pluginRef: prefix-cache-scorer, weight: 3}
+      - {pluginRef: queue-scorer, weight: 1}
+      - {pluginRef: kv-cache-utilization-scorer, weight: 1}
+      - {pluginRef: max-score-picker}
+  - name: prefill
+    plugins:
+      - {pluginRef: prefill-filter}
+      - {pluginRef: queue-scorer, weight: 1}
+      - {pluginRef: max-score-picker}
+"""
+
+
+@dataclass
+class NodeConfig:
+    model: ModelConfig
+    rank: int = 0
+    world_size: int = 1
+    topology: str = "mono"
+    epp_yaml: str = ""
+    device: str = "cpu"
+    dtype: Any = torch.bfloat16
+    kv_blocks: Optional[int] = None
+    kv_budget_bytes: int = 8 << 30
+    prefill_chunk_tokens: int = 8192
+    max_decode_batch: int = 256
+    flow_control: bool = False
+    fc_bands: List[BandConfig] = field(default_factory=list)
+    fc_global_max_items: Optional[int] = None
+    route_batch_per_step: int = 64
+    seed: int = 0
+    mailbox_group: Any = None
+    transfer_group: Any = None
+
+
+@dataclass
+class Completion:
+    request_id: str
+    usage: Usage
+    tokens: List[int] = field(default_factory=list)
+    error: str = ""
+
+
+class NodeRunner:
+    def __init__(self, cfg: NodeConfig):
+        self.cfg = cfg
+        self.rank = cfg.rank
+        self.topology = NodeTopology.parse(cfg.topology, cfg.world_size)
+        self.my_spec = self.topology.ranks[self.rank]
+        self.mailbox = Mailbox(cfg.mailbox_group, self.rank, cfg.world_size)
+
+        self.engine = EngineWorker(
+            cfg.model, cfg.device, role=self.my_spec.role,
+            kv_blocks=cfg.kv_blocks, kv_budget_bytes=cfg.kv_budget_bytes,
+            dtype=cfg.dtype, prefill_chunk_tokens=cfg.prefill_chunk_tokens,
+            max_decode_batch=cfg.max_decode_batch, seed=cfg.seed)
+        self.transfer = KVTransferEngine(self.engine.pool.tensor, self.rank,
+                                         group=cfg.transfer_group)
+        self._outbox: List[Dict[str, Any]] = []
+        # decode-side: requests waiting for a remote prefill's KV
+        self._pending_adoption: Dict[str, Dict[str, Any]] = {}
+        # prefill-side: req_id -> decode rank for the eventual hand-off
+        self._handoff_dst: Dict[str, int] = {}
+        self._step = 0
+
+        self.is_router = (self.rank == 0)
+        if self.is_router:
+            self._init_router()
+
+    # ------------------------------------------------------------------
+    def _init_router(self) -> None:
+        cfg = self.cfg
+        self.loaded: LoadedConfig = load_config(cfg.epp_yaml or
+                                                DEFAULT_EPP_CONFIG)
+        self.datastore = Datastore()
+        for spec in self.topology.ranks:
+            ep = make_endpoint(f"gpu{spec.rank}", spec.rank, rank=spec.rank,
+                               role=spec.role_label,
+                               address=f"rank:{spec.rank}")
+            ep.update_metrics(Metrics(cache_num_blocks=self.engine.pool.num_blocks,
+                                      cache_block_size=self.engine.pool.block_size))
+            self.datastore.add_endpoint(ep)
+        self.datastore.set_pool_ready()
+        self.detector = UtilizationSaturationDetector()
+        candidates = EndpointCandidates(self.datastore, cache_ttl_s=0.0)
+        if cfg.flow_control or self.loaded.gate("flowControl"):
+            bands = cfg.fc_bands or [BandConfig(0), BandConfig(-1)]
+            registry = FlowRegistry(bands=bands,
+                                    global_max_items=cfg.fc_global_max_items)
+            self.flow = FlowController(
+                registry, lambda item: True,
+                saturated_fn=lambda: self.detector.is_saturated(
+                    candidates.all()))
+            self.flow.start()
+            admission = FlowControlAdmissionController(self.flow)
+        else:
+            self.flow = None
+            admission = LegacyAdmissionController(self.detector)
+        self.director = Director(
+            datastore=self.datastore,
+            scheduler=Scheduler(self.loaded.scheduler_config),
+            admission=admission, candidates=candidates,
+            config=self.loaded.request_control)
+        self._arrivals: List[LLMRequest] = []
+        self._decisions: Dict[str, RoutingDecision] = {}
+        self._completions: List[Completion] = []
+        self._assign_seq = 0
+        # flow-control mode: admission blocks in the queue, so routing runs
+        # on a pool and finished decisions drain into the outbox each step
+        self._route_pool = (ThreadPoolExecutor(max_workers=64,
+                                               thread_name_prefix="route")
+                            if self.flow is not None else None)
+        self._routed: "queue_mod.Queue" = queue_mod.Queue()
+
+    # ------------------------------------------------------------------
+    # rank-0 API
+    def submit(self, req: LLMRequest) -> None:
+        assert self.is_router
+        self._arrivals.append(req)
+
+    def drain_completions(self) -> List[Completion]:
+        assert self.is_router
+        out = self._completions
+        self._completions = []
+        return out
+
+    @property
+    def inflight(self) -> int:
+        return len(self._decisions) if self.is_router else 0
+
+    # ------------------------------------------------------------------
+    def step(self) -> None:
+        """One lockstep node iteration on every rank."""
+        if self.is_router:
+            self._route_arrivals()
+        self._outbox.append({"type": "metrics", "src": self.rank,
+                             "m": self._metrics_payload()})
+        msgs = self.mailbox.exchange(self._outbox)
+        self._outbox = []
+        self._process_messages(msgs)
+        self._execute_transfers(msgs)
+        outputs = self.engine.step()
+        self._handle_outputs(outputs)
+        self._step += 1
+
+    # ---- router side ----
+    def _route_arrivals(self) -> None:
+        if self._route_pool is not None:
+            # hand all arrivals to the pool; drain whatever finished routing
+            for req in self._arrivals:
+                self._route_pool.submit(self._route_one_threaded, req)
+            self._arrivals = []
+            while True:
+                try:
+                    kind, req, payload = self._routed.get_nowait()
+                except queue_mod.Empty:
+                    break
+                if kind == "deny":
+                    prom.request_error_total.labels(req.model,
+                                                    payload.reason).inc()
+                    self._completions.append(Completion(
+                        request_id=req.request_id, usage=Usage(),
+                        error=payload.reason))
+                else:
+                    self._emit_assignment(req, payload)
+            return
+        n = min(len(self._arrivals), self.cfg.route_batch_per_step)
+        batch, self._arrivals = self._arrivals[:n], self._arrivals[n:]
+        for req in batch:
+            try:
+                decision = self.director.handle_request(req)
+            except AdmissionDenied as e:
+                prom.request_error_total.labels(req.model, e.reason).inc()
+                self._completions.append(Completion(
+                    request_id=req.request_id, usage=Usage(),
+                    error=e.reason))
+                continue
+            self._emit_assignment(req, decision)
+
+    def _route_one_threaded(self, req: LLMRequest) -> None:
+        try:
+            decision = self.director.handle_request(req)
+            self._routed.put(("ok", req, decision))
+        except AdmissionDenied as e:
+            self._routed.put(("deny", req, e))
+        except Exception as e:  # pragma: no cover
+            log.error("routing failed", err=str(e))
+            self._routed.put(("deny", req,
+                              AdmissionDenied("internal", str(e), 500)))
+
+    def _emit_assignment(self, req: LLMRequest,
+                         decision: RoutingDecision) -> None:
+            self._decisions[req.request_id] = decision
+            decode_rank = decision.target.metadata.rank
+            msg = {"type": "assign", "req_id": req.request_id,
+                   "dst": decode_rank,
+                   "tokens": req.prompt_tokens or [],
+                   "max_tokens": req.max_tokens,
+                   "temperature": req.temperature,
+                   "is_embedding": req.is_embedding,
+                   "cached": self._cached_tokens(decision, decode_rank),
+                   "seq": self._assign_seq}
+            self._assign_seq += 1
+            prefill_hdr = req.headers.get("x-prefiller-host-port")
+            if prefill_hdr:
+                msg["prefill"] = int(prefill_hdr.split(":")[-1])
+            encode_hdr = req.headers.get("x-encoder-hosts-ports")
+            if encode_hdr:
+                msg["encode"] = [int(h.split(":")[-1])
+                                 for h in encode_hdr.split(",")]
+            self._outbox.append(msg)
+
+    def _cached_tokens(self, decision: RoutingDecision,
+                       decode_rank: int) -> int:
+        info = decision.ctx.attributes.get("prefix.PrefixCacheMatchInfo")
+        if info is None:
+            return 0
+        return info.match_blocks.get(f"gpu{decode_rank}", 0) * \
+            info.block_size_tokens
+
+    # ---- worker side ----
+    def _metrics_payload(self) -> Dict[str, Any]:
+        m = self.engine.metrics_snapshot()
+        return {"q": m.waiting_queue_size, "r": m.running_requests_size,
+                "kv": m.kv_cache_usage, "nb": m.cache_num_blocks,
+                "bs": m.cache_block_size}
+
+    def _process_messages(self, msgs: List[Dict[str, Any]]) -> None:
+        for m in msgs:
+            t = m.get("type")
+            if t == "metrics" and self.is_router:
+                ep = self.datastore.get_endpoint(f"gpu{m['src']}")
+                if ep is not None:
+                    ep.update_metrics(Metrics(
+                        waiting_queue_size=m["m"]["q"],
+                        running_requests_size=m["m"]["r"],
+                        kv_cache_usage=m["m"]["kv"],
+                        cache_num_blocks=m["m"]["nb"],
+                        cache_block_size=m["m"]["bs"]))
+            elif t == "assign":
+                self._handle_assign(m)
+            elif t == "done" and self.is_router:
+                self._handle_done(m)
+
+    def _handle_assign(self, m: Dict[str, Any]) -> None:
+        prefill_rank = m.get("prefill")
+        decode_rank = m["dst"]
+        req = EngineRequest(
+            request_id=m["req_id"], prompt_tokens=list(m["tokens"]),
+            max_tokens=m["max_tokens"], temperature=m["temperature"],
+            is_embedding=m.get("is_embedding", False),
+            cached_tokens=m.get("cached", 0))
+        if prefill_rank is not None and prefill_rank != decode_rank:
+            if self.rank == prefill_rank:
+                req.prefill_only = True
+                self._handoff_dst[m["req_id"]] = decode_rank
+                self.engine.add_request(req)
+            elif self.rank == decode_rank:
+                self._pending_adoption[m["req_id"]] = {"req": req,
+                                                       "src": prefill_rank}
+        elif self.rank == decode_rank:
+            self.engine.add_request(req)
+
+    def _handle_done(self, m: Dict[str, Any]) -> None:
+        decision = self._decisions.pop(m["req_id"], None)
+        usage = Usage(prompt_tokens=m.get("prompt_tokens", 0),
+                      completion_tokens=m.get("completion_tokens", 0),
+                      cached_tokens=m.get("cached_tokens", 0),
+                      ttft_ms=m.get("ttft_ms"), tpot_ms=m.get("tpot_ms"),
+                      e2e_ms=m.get("e2e_ms"))
+        if decision is not None:
+            self.director.handle_response_complete(decision, usage)
+        self._completions.append(Completion(
+            request_id=m["req_id"], usage=usage,
+            tokens=m.get("tokens", []), error=m.get("error", "")))
+
+    # ---- transfers (the NIXL-v2 step 2/3 replacement) ----
+    def _execute_transfers(self, msgs: List[Dict[str, Any]]) -> None:
+        jobs = sorted((m for m in msgs if m.get("type") == "kv_ready"),
+                      key=lambda m: m["req_id"])
+        for job in jobs:
+            src, dst = job["src"], job["dst"]
+            if self.rank == src:
+                self.transfer.send_blocks(dst, job["blocks"])
+                self.engine.release_prefilled(job["req_id"])
+            elif self.rank == dst:
+                n = len(job["blocks"])
+                local = self.engine.mgr.take_blocks(n)
+                if local is None:
+                    # keep the P2P matched: receive into scratch and drop
+                    scratch = self.transfer._staging(n)
+                    torch.distributed.recv(scratch, src=src,
+                                           group=self.cfg.transfer_group)
+                    self._outbox.append({"type": "done",
+                                         "req_id": job["req_id"],
+                                         "error": "kv_exhausted"})
+                    self._pending_adoption.pop(job["req_id"], None)
+                    continue
+                self.transfer.recv_blocks(src, local)
+                pending = self._pending_adoption.pop(job["req_id"], None)
+                if pending is None:
+                    self.engine.mgr.release_blocks(local)
+                    continue
+                self.engine.admit_transferred(pending["req"], local,
+                                              job["seq_len"],
+                                              job["first_token"])
+
+    # ---- engine outputs -> messages ----
+    def _handle_outputs(self, outputs: List[RequestOutput]) -> None:
+        for out in outputs:
+            if out.kind == "prefill_done":
+                # announce KV availability (next step's exchange)
+                decode_rank = self._handoff_dst.pop(out.request_id, 0)
+                self._outbox.append({
+                    "type": "kv_ready", "req_id": out.request_id,
+                    "src": self.rank, "dst": decode_rank,
+                    "blocks": out.kv_blocks, "seq_len": out.seq_len,
+                    "first_token": out.first_token,
+                    "ttft_ms": out.ttft_ms})
+            elif out.kind == "embedding" and out.finished:
+                self._emit_done(out, tokens=[])
+            elif out.finished:
+                self._emit_done(out, tokens=None)
+
+    def _emit_done(self, out: RequestOutput, tokens) -> None:
+        msg = {"type": "done", "req_id": out.request_id,
+               "prompt_tokens": out.prompt_tokens,
+               "completion_tokens": out.completion_tokens,
+               "cached_tokens": out.cached_tokens,
+               "tokens": out.all_tokens or [],
+               "ttft_ms": out.ttft_ms, "tpot_ms": out.tpot_ms,
+               "e2e_ms": out.e2e_ms}
+        if self.is_router:
+            self._handle_done(msg)
+        else:
+            self._outbox.append(msg)
+
+    # ------------------------------------------------------------------
+    def shutdown(self) -> None:
+        if self.is_router and self.flow is not None:
+            self.flow.stop()

@@ -1,0 +1,89 @@
+"""Op dispatch: gfx950 HIP kernels on GPU tensors, torch reference on CPU.
+
+Policy (driver contract): on a GPU box the HIP extension is REQUIRED — a
+CUDA-device tensor with `_hip_ops` missing raises immediately rather than
+silently falling back to eager torch. CPU tensors (tests, tiny gloo runs)
+use ops.ref.
+"""
+from typing import Optional
+
+import torch
+
+from . import ref
+
+try:
+    from .. import _hip_ops as _ext
+except ImportError:  # extension not built
+    _ext = None
+
+
+def hip_ops():
+    """The raw extension module (router prefix kernels etc.). Fail-loud."""
+    if _ext is None:
+        raise RuntimeError(
+            "_hip_ops extension is not built; run "
+            "`python setup.py build_ext --inplace` (gfx950 HIP kernels are "
+            "mandatory on GPU hosts — no eager fallback)")
+    return _ext
+
+
+def _use_hip(t: torch.Tensor) -> bool:
+    if not t.is_cuda:
+        return False
+    hip_ops()  # raises if missing on a GPU path
+    return True
+
+
+def rmsnorm(x: torch.Tensor, w: torch.Tensor, eps: float,
+            residual: Optional[torch.Tensor] = None) -> torch.Tensor:
+    if _use_hip(x):
+        return _ext.rmsnorm(x, w, eps, residual)
+    return ref.rmsnorm(x, w, eps, residual)
+
+
+def rope(q: torch.Tensor, k: torch.Tensor, cos_sin: torch.Tensor,
+         positions: torch.Tensor) -> None:
+    if _use_hip(q):
+        _ext.rope(q, k, cos_sin, positions.to(torch.int32))
+        return
+    ref.rope(q, k, cos_sin, positions)
+
+
+def silu_mul(gate_up: torch.Tensor) -> torch.Tensor:
+    if _use_hip(gate_up):
+        return _ext.silu_mul(gate_up)
+    return ref.silu_mul(gate_up)
+
+
+def reshape_and_cache(k_new: torch.Tensor, v_new: torch.Tensor,
+                      k_cache: torch.Tensor, v_cache: torch.Tensor,
+                      slots: torch.Tensor) -> None:
+    if _use_hip(k_new):
+        _ext.reshape_and_cache(k_new, v_new, k_cache, v_cache,
+                               slots.to(torch.int64))
+        return
+    ref.reshape_and_cache(k_new, v_new, k_cache, v_cache, slots)
+
+
+def paged_attention(q: torch.Tensor, k_cache: torch.Tensor,
+                    v_cache: torch.Tensor, block_tables: torch.Tensor,
+                    seq_lens: torch.Tensor, scale: float) -> torch.Tensor:
+    if _use_hip(q):
+        return _ext.paged_attention(q, k_cache, v_cache,
+                                    block_tables.to(torch.int32),
+                                    seq_lens.to(torch.int32), scale)
+    return ref.paged_attention(q, k_cache, v_cache, block_tables, seq_lens,
+                               scale)
+
+
+def move_blocks(pool: torch.Tensor, staging: torch.Tensor,
+                block_ids: torch.Tensor, is_scatter: bool) -> None:
+    if _use_hip(pool):
+        _ext.move_blocks(pool, staging, block_ids.to(torch.int32), is_scatter)
+        return
+    # CPU reference: pool [L, 2, NB, KVH, BS, D]; staging [n, L, 2, KVH, BS, D]
+    ids = block_ids.long()
+    if is_scatter:
+        pool[:, :, ids] = staging.permute(1, 2, 0, 3, 4, 5)
+    else:
+        staging.copy_(pool[:, :, ids].permute(2, 0, 1, 3, 4, 5))

@@ -9,6 +9,7 @@ candidate location (subset hint) -> DataProducer plugins (DAG-ordered,
 prepare (target endpoint + stage headers) -> PreRequest plugins.
 """
 import random
+import threading
 import time
 from dataclasses import dataclass, field
 from typing import Dict, List, Optional
@@ -100,6 +101,10 @@ class Director:
         self.config = config
         self.config.topo_sort_producers()
         self._rng = random.Random(seed)
+        # serializes produce+schedule+prepare when requests are routed from
+        # several threads (flow-control mode); admission itself (which may
+        # block in the queue) stays outside the lock.
+        self._sched_lock = threading.Lock()
 
     # ---- request path ----
     def handle_request(self, req: LLMRequest) -> RoutingDecision:
@@ -123,22 +128,23 @@ class Director:
             # admission may block (flow control) or shed
             self.admission.admit(ctx, endpoints)
 
-            self._run_producers(ctx, endpoints)
-            for admitter in self.config.admitters:
-                ok, reason = admitter.admit(ctx, endpoints)
-                if not ok:
-                    raise AdmissionDenied("admitter_denied", reason)
+            with self._sched_lock:
+                self._run_producers(ctx, endpoints)
+                for admitter in self.config.admitters:
+                    ok, reason = admitter.admit(ctx, endpoints)
+                    if not ok:
+                        raise AdmissionDenied("admitter_denied", reason)
 
-            t_sched = time.monotonic()
-            result = self.scheduler.schedule(ctx, endpoints)
-            prom.scheduler_e2e.observe(time.monotonic() - t_sched)
+                t_sched = time.monotonic()
+                result = self.scheduler.schedule(ctx, endpoints)
+                prom.scheduler_e2e.observe(time.monotonic() - t_sched)
 
-            target = result.primary.target if result.primary else None
-            if target is None:
-                raise AdmissionDenied("no_target",
-                                      "scheduler produced no target",
-                                      status=503)
-            self._prepare_request(ctx, result, target)
+                target = result.primary.target if result.primary else None
+                if target is None:
+                    raise AdmissionDenied("no_target",
+                                          "scheduler produced no target",
+                                          status=503)
+                self._prepare_request(ctx, result, target)
             latency_ms = (time.monotonic() - t0) * 1e3
             prom.request_total.labels(req.model, req.target_model).inc()
             prom.running_requests.labels(req.target_model).inc()

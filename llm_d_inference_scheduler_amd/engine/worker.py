@@ -33,12 +33,13 @@ class EngineRequest:
     # runtime state
     computed: int = 0               # prompt tokens already prefilled
     generated: List[int] = field(default_factory=list)
+    slo_ok: bool = True             # TTFT met the SLO (goodput accounting)
     arrival_t: float = 0.0
     first_token_t: float = 0.0
 
     def __post_init__(self):
         if not self.arrival_t:
-            self.arrival_t = time.monotonic()
+            self.arrival_t = time.time()
 
     @property
     def prompt_len(self) -> int:
@@ -76,6 +77,7 @@ class EngineWorker:
                  prefill_chunk_tokens: int = 8192,
                  max_decode_batch: int = 256,
                  max_model_len: int = 8192,
+                 ttft_slo_ms: float = None,
                  seed: int = 0):
         self.cfg = config
         self.device = torch.device(device)
@@ -84,6 +86,7 @@ class EngineWorker:
         self.prefill_chunk_tokens = prefill_chunk_tokens
         self.max_decode_batch = max_decode_batch
         self.max_model_len = max_model_len
+        self.ttft_slo_ms = ttft_slo_ms
         if kv_blocks is None:
             kv_blocks = KVPool.blocks_for_budget(config, kv_budget_bytes,
                                                  dtype_bytes=2)
@@ -95,6 +98,7 @@ class EngineWorker:
         self._by_id: Dict[str, EngineRequest] = {}
         self.steps = 0
         self.total_generated = 0
+        self.total_generated_slo = 0
         self.total_prefilled = 0
 
     # ------------------------------------------------------------------
@@ -111,7 +115,10 @@ class EngineWorker:
         req.computed = req.prompt_len
         req.generated = [first_token]
         if not req.first_token_t:
-            req.first_token_t = time.monotonic()
+            req.first_token_t = time.time()
+        if self.ttft_slo_ms is not None and req.arrival_t and \
+                (req.first_token_t - req.arrival_t) * 1e3 > self.ttft_slo_ms:
+            req.slo_ok = False
         self._by_id[req.request_id] = req
         self.running.append(req)
 
@@ -210,7 +217,7 @@ class EngineWorker:
             batch, self.pool.tensor, embeddings_out=bool(embedding_reqs))
 
         outputs: List[RequestOutput] = []
-        now = time.monotonic()
+        now = time.time()
         if embedding_reqs:
             # embeddings: mean-pool each finished sequence's chunk rows
             for i, (req, chunk) in enumerate(selected):
@@ -241,9 +248,12 @@ class EngineWorker:
 
     def _finish_prefills(self, finishing, logits, outputs) -> None:
         tokens = self._sample(logits, [r.temperature for r in finishing])
-        now = time.monotonic()
+        now = time.time()
         for req, tok in zip(finishing, tokens):
             req.first_token_t = now
+            if self.ttft_slo_ms is not None and req.arrival_t and \
+                    (now - req.arrival_t) * 1e3 > self.ttft_slo_ms:
+                req.slo_ok = False
             self.waiting.remove(req)
             if req.prefill_only:
                 outputs.append(RequestOutput(
@@ -258,6 +268,8 @@ class EngineWorker:
                 self._by_id.pop(req.request_id, None)
             else:
                 req.generated.append(int(tok))
+                if req.slo_ok:
+                    self.total_generated_slo += 1
                 self.running.append(req)
                 outputs.append(RequestOutput(
                     request_id=req.request_id, new_tokens=[int(tok)],
@@ -304,10 +316,12 @@ class EngineWorker:
         logits = self.model.forward(batch, self.pool.tensor)
         tokens = self._sample(logits, [r.temperature for r in active])
         outputs: List[RequestOutput] = []
-        now = time.monotonic()
+        now = time.time()
         for req, tok in zip(active, tokens):
             req.generated.append(int(tok))
             self.total_generated += 1
+            if req.slo_ok:
+                self.total_generated_slo += 1
             finished = len(req.generated) >= req.max_tokens
             out = RequestOutput(request_id=req.request_id,
                                 new_tokens=[int(tok)], finished=finished)

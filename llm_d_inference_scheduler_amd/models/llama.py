@@ -50,13 +50,25 @@ class LlamaRunner:
         self.device = device
         self.dtype = dtype
         self.scale = 1.0 / math.sqrt(config.head_dim)
-        gen = torch.Generator(device="cpu").manual_seed(seed)
         c = config
+        if self.device.type == "cuda":
+            # init directly on the GPU (16 GB of weights for Llama-3-8B —
+            # CPU-side init would dominate startup). Same seed => identical
+            # weights on every rank (required by the P/D KV hand-off).
+            torch.cuda.manual_seed_all(seed)
 
-        def w(*shape, std=0.02):
-            t = torch.empty(*shape, dtype=torch.float32)
-            t.normal_(0.0, std, generator=gen)
-            return t.to(dtype).to(device)
+            def w(*shape, std=0.02):
+                t = torch.empty(*shape, dtype=torch.float32,
+                                device=self.device)
+                t.normal_(0.0, std)
+                return t.to(dtype)
+        else:
+            gen = torch.Generator(device="cpu").manual_seed(seed)
+
+            def w(*shape, std=0.02):
+                t = torch.empty(*shape, dtype=torch.float32)
+                t.normal_(0.0, std, generator=gen)
+                return t.to(dtype).to(device)
 
         out_std = 0.02 / math.sqrt(2 * c.num_layers)
         self.embed = w(c.vocab_size, c.hidden_size)

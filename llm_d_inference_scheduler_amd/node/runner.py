@@ -88,6 +88,7 @@ class NodeConfig:
     fc_bands: List[BandConfig] = field(default_factory=list)
     fc_global_max_items: Optional[int] = None
     route_batch_per_step: int = 64
+    ttft_slo_ms: Optional[float] = None
     seed: int = 0
     mailbox_group: Any = None
     transfer_group: Any = None
@@ -113,7 +114,8 @@ class NodeRunner:
             cfg.model, cfg.device, role=self.my_spec.role,
             kv_blocks=cfg.kv_blocks, kv_budget_bytes=cfg.kv_budget_bytes,
             dtype=cfg.dtype, prefill_chunk_tokens=cfg.prefill_chunk_tokens,
-            max_decode_batch=cfg.max_decode_batch, seed=cfg.seed)
+            max_decode_batch=cfg.max_decode_batch,
+            ttft_slo_ms=cfg.ttft_slo_ms, seed=cfg.seed)
         self.transfer = KVTransferEngine(self.engine.pool.tensor, self.rank,
                                          group=cfg.transfer_group)
         self._outbox: List[Dict[str, Any]] = []
@@ -165,6 +167,8 @@ class NodeRunner:
         self._decisions: Dict[str, RoutingDecision] = {}
         self._completions: List[Completion] = []
         self._assign_seq = 0
+        from collections import deque
+        self.epp_latencies = deque(maxlen=100_000)  # ms, per routed request
         # flow-control mode: admission blocks in the queue, so routing runs
         # on a pool and finished decisions drain into the outbox each step
         self._route_pool = (ThreadPoolExecutor(max_workers=64,
@@ -176,6 +180,7 @@ class NodeRunner:
     # rank-0 API
     def submit(self, req: LLMRequest) -> None:
         assert self.is_router
+        req.headers.setdefault("x-arrival-wall", str(time.time()))
         self._arrivals.append(req)
 
     def drain_completions(self) -> List[Completion]:
@@ -251,6 +256,7 @@ class NodeRunner:
     def _emit_assignment(self, req: LLMRequest,
                          decision: RoutingDecision) -> None:
             self._decisions[req.request_id] = decision
+            self.epp_latencies.append(decision.epp_latency_ms)
             decode_rank = decision.target.metadata.rank
             msg = {"type": "assign", "req_id": req.request_id,
                    "dst": decode_rank,
@@ -259,6 +265,7 @@ class NodeRunner:
                    "temperature": req.temperature,
                    "is_embedding": req.is_embedding,
                    "cached": self._cached_tokens(decision, decode_rank),
+                   "arrival": float(req.headers.get("x-arrival-wall", 0) or 0),
                    "seq": self._assign_seq}
             self._assign_seq += 1
             prefill_hdr = req.headers.get("x-prefiller-host-port")
@@ -309,7 +316,8 @@ class NodeRunner:
             request_id=m["req_id"], prompt_tokens=list(m["tokens"]),
             max_tokens=m["max_tokens"], temperature=m["temperature"],
             is_embedding=m.get("is_embedding", False),
-            cached_tokens=m.get("cached", 0))
+            cached_tokens=m.get("cached", 0),
+            arrival_t=m.get("arrival") or 0.0)
         if prefill_rank is not None and prefill_rank != decode_rank:
             if self.rank == prefill_rank:
                 req.prefill_only = True

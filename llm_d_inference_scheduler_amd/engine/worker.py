@@ -105,6 +105,12 @@ class EngineWorker:
     def add_request(self, req: EngineRequest) -> None:
         if len(req.prompt_tokens) >= self.max_model_len:
             req.prompt_tokens = req.prompt_tokens[:self.max_model_len - 1]
+        if any(t >= self.cfg.vocab_size or t < 0
+               for t in req.prompt_tokens):
+            # defensive clamp: a mismatched tokenizer must not crash the
+            # engine loop (ids fold into the vocab deterministically)
+            req.prompt_tokens = [t % self.cfg.vocab_size
+                                 for t in req.prompt_tokens]
         self.waiting.append(req)
         self._by_id[req.request_id] = req
 

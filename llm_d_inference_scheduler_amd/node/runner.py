@@ -123,6 +123,9 @@ class NodeRunner:
         self._pending_adoption: Dict[str, Dict[str, Any]] = {}
         # prefill-side: req_id -> decode rank for the eventual hand-off
         self._handoff_dst: Dict[str, int] = {}
+        # decode-side: req_ids whose tokens stream back per step
+        self._streaming_ids: set = set()
+        self._token_events: List[tuple] = []  # rank0: (req_id, [tokens])
         self._step = 0
 
         self.is_router = (self.rank == 0)
@@ -143,6 +146,12 @@ class NodeRunner:
                                       cache_block_size=self.engine.pool.block_size))
             self.datastore.add_endpoint(ep)
         self.datastore.set_pool_ready()
+        # the shared in-process tokenizer must match the model's vocab
+        from ..models.tokenizer import HashTokenizer
+        from ..plugins.producers import TokenProducer
+        for p in self.loaded.plugins.values():
+            if isinstance(p, TokenProducer):
+                p.tokenizer = HashTokenizer(cfg.model.vocab_size)
         self.detector = UtilizationSaturationDetector()
         candidates = EndpointCandidates(self.datastore, cache_ttl_s=0.0)
         if cfg.flow_control or self.loaded.gate("flowControl"):
@@ -163,6 +172,17 @@ class NodeRunner:
             scheduler=Scheduler(self.loaded.scheduler_config),
             admission=admission, candidates=candidates,
             config=self.loaded.request_control)
+        # batched gfx950 prefix path: hash+match the whole admission batch
+        # in two kernel launches on the router rank's GPU
+        self._approx = None
+        self._gpu_prefix = None
+        for p in self.loaded.plugins.values():
+            if isinstance(p, ApproxPrefixCacheProducer):
+                self._approx = p
+        if self._approx is not None and str(cfg.device).startswith("cuda"):
+            from ..ops.prefix import GpuPrefixIndex
+            self._gpu_prefix = GpuPrefixIndex(cfg.device)
+            self._approx.attach_gpu_index(self._gpu_prefix)
         self._arrivals: List[LLMRequest] = []
         self._decisions: Dict[str, RoutingDecision] = {}
         self._completions: List[Completion] = []
@@ -187,6 +207,12 @@ class NodeRunner:
         assert self.is_router
         out = self._completions
         self._completions = []
+        return out
+
+    def drain_token_events(self) -> List[tuple]:
+        assert self.is_router
+        out = self._token_events
+        self._token_events = []
         return out
 
     @property
@@ -231,9 +257,11 @@ class NodeRunner:
             return
         n = min(len(self._arrivals), self.cfg.route_batch_per_step)
         batch, self._arrivals = self._arrivals[:n], self._arrivals[n:]
-        for req in batch:
+        precomputed = self._gpu_prefix_batch(batch)
+        for i, req in enumerate(batch):
             try:
-                decision = self.director.handle_request(req)
+                decision = self.director.handle_request(
+                    req, precomputed[i] if precomputed else None)
             except AdmissionDenied as e:
                 prom.request_error_total.labels(req.model, e.reason).inc()
                 self._completions.append(Completion(
@@ -241,6 +269,36 @@ class NodeRunner:
                     error=e.reason))
                 continue
             self._emit_assignment(req, decision)
+
+    def _gpu_prefix_batch(self, batch: List[LLMRequest]):
+        """One hash_prompts + one match_longest launch for the whole
+        admission batch (SURVEY.md §2.6 MI355X mapping). Returns per-request
+        precomputed-attribute dicts, or None to use the host path."""
+        if (self._gpu_prefix is None or len(batch) < 2 or
+                any(not r.prompt_tokens for r in batch)):
+            return None
+        from ..datalayer.attributes import (PREFIX_CACHE_MATCH_INFO as KEY,
+                                            PrefixCacheMatchInfo)
+        ap = self._approx
+        seed0 = ap._seed(batch[0].target_model or batch[0].model)
+        hashes_dev, counts_dev = self._gpu_prefix.hash_prompts_batch(
+            [r.prompt_tokens for r in batch], ap.block_size, ap.max_blocks,
+            seed0)
+        n_eps = len(self.topology.ranks)
+        match = self._gpu_prefix.match_batch(hashes_dev, counts_dev, n_eps)
+        match = match.cpu().numpy()
+        counts = counts_dev.cpu().numpy()
+        hashes = hashes_dev.cpu().numpy()
+        out = []
+        for i in range(len(batch)):
+            info = PrefixCacheMatchInfo(total_blocks=int(counts[i]),
+                                        block_size_tokens=ap.block_size)
+            for e in range(n_eps):
+                info.match_blocks[f"gpu{e}"] = int(match[i, e])
+            out.append({KEY: info,
+                        "_state:prefix_hashes":
+                            hashes[i, :counts[i]].astype("uint64")})
+        return out
 
     def _route_one_threaded(self, req: LLMRequest) -> None:
         try:
@@ -264,6 +322,7 @@ class NodeRunner:
                    "max_tokens": req.max_tokens,
                    "temperature": req.temperature,
                    "is_embedding": req.is_embedding,
+                   "stream": req.streaming,
                    "cached": self._cached_tokens(decision, decode_rank),
                    "arrival": float(req.headers.get("x-arrival-wall", 0) or 0),
                    "seq": self._assign_seq}
@@ -308,6 +367,11 @@ class NodeRunner:
                 self._handle_assign(m)
             elif t == "done" and self.is_router:
                 self._handle_done(m)
+            elif t == "tokens" and self.is_router:
+                self._token_events.append((m["req_id"], m["toks"]))
+                decision = self._decisions.get(m["req_id"])
+                if decision is not None:
+                    self.director.handle_response_chunk(decision, m["toks"])
 
     def _handle_assign(self, m: Dict[str, Any]) -> None:
         prefill_rank = m.get("prefill")
@@ -328,6 +392,8 @@ class NodeRunner:
                                                        "src": prefill_rank}
         elif self.rank == decode_rank:
             self.engine.add_request(req)
+        if m.get("stream") and self.rank == decode_rank:
+            self._streaming_ids.add(m["req_id"])
 
     def _handle_done(self, m: Dict[str, Any]) -> None:
         decision = self._decisions.pop(m["req_id"], None)
@@ -387,8 +453,18 @@ class NodeRunner:
                     "ttft_ms": out.ttft_ms})
             elif out.kind == "embedding" and out.finished:
                 self._emit_done(out, tokens=[])
-            elif out.finished:
-                self._emit_done(out, tokens=None)
+            else:
+                if out.request_id in self._streaming_ids and out.new_tokens:
+                    msg = {"type": "tokens", "req_id": out.request_id,
+                           "toks": list(out.new_tokens)}
+                    if self.is_router:
+                        self._token_events.append((out.request_id,
+                                                   msg["toks"]))
+                    else:
+                        self._outbox.append(msg)
+                if out.finished:
+                    self._streaming_ids.discard(out.request_id)
+                    self._emit_done(out, tokens=None)
 
     def _emit_done(self, out: RequestOutput, tokens) -> None:
         msg = {"type": "done", "req_id": out.request_id,

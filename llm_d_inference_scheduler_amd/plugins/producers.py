@@ -119,6 +119,8 @@ class ApproxPrefixCacheProducer(DataProducer):
 
     def produce(self, ctx: SchedulingContext,
                 endpoints: List[Endpoint]) -> None:
+        if PREFIX_CACHE_MATCH_INFO in ctx.attributes:
+            return  # precomputed by the batched gfx950 admission path
         if not self._auto_tuned:
             self._auto_tune(endpoints)
         hashes = self.hash_request(ctx)

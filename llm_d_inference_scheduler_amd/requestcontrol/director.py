@@ -107,12 +107,23 @@ class Director:
         self._sched_lock = threading.Lock()
 
     # ---- request path ----
-    def handle_request(self, req: LLMRequest) -> RoutingDecision:
+    def handle_request(self, req: LLMRequest,
+                       precomputed: Optional[Dict] = None) -> RoutingDecision:
+        """`precomputed`: request-scoped attributes produced ahead of this
+        call (the batched gfx950 prefix hash/match over a whole admission
+        batch) — producers skip keys already present."""
         t0 = time.monotonic()
         tracer = get_tracer()
         with tracer.span("director.handle_request",
                          request_id=req.request_id, model=req.model):
             ctx = SchedulingContext(request=req)
+            if precomputed:
+                ctx.attributes.update(
+                    {k: v for k, v in precomputed.items()
+                     if not k.startswith("_state:")})
+                ctx.state.update(
+                    {k[7:]: v for k, v in precomputed.items()
+                     if k.startswith("_state:")})
             self._mutate_model(req)
             self._resolve_objective(req)
 

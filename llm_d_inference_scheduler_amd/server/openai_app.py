@@ -1,0 +1,151 @@
+"""OpenAI-compatible HTTP front door (FastAPI).
+
+Parity surface: the request paths the reference's parsers handle
+(/v1/completions, /v1/chat/completions, /v1/embeddings, /v1/responses —
+parsers/openai/openai.go) plus /metrics (Prometheus, vLLM-compatible
+series) and /healthz (health.go:52 readiness semantics: pool synced).
+Errors surface the canonical `x-request-dropped-reason` header
+(pkg/common/error). SSE streaming re-emits per-token chunks like the
+sidecar's chunked decode path (decode.go SSE re-emission).
+"""
+import asyncio
+import json
+import time
+import uuid
+from typing import Optional
+
+from fastapi import FastAPI, Request, Response
+from fastapi.responses import JSONResponse, StreamingResponse
+
+from ..handlers.parsers import ParserMux
+from ..metrics import prom
+from ..models.tokenizer import HashTokenizer
+from ..scheduling.types import LLMRequest
+from .service import NodeService
+
+DROPPED_REASON_HEADER = "x-request-dropped-reason"
+
+
+def build_app(service: NodeService,
+              tokenizer: Optional[HashTokenizer] = None) -> FastAPI:
+    app = FastAPI(title="llm-d-inference-scheduler-amd")
+    mux = ParserMux()
+    tok = tokenizer or HashTokenizer()
+
+    def _error(status: int, reason: str, detail: str = "") -> JSONResponse:
+        return JSONResponse(
+            status_code=status,
+            headers={DROPPED_REASON_HEADER: reason},
+            content={"error": {"message": detail or reason,
+                               "type": reason, "code": status}})
+
+    async def _handle(request: Request, path: str):
+        body = await request.body()
+        headers = {k.lower(): v for k, v in request.headers.items()}
+        result = mux.parse_request(body, headers, path)
+        if result.error:
+            return _error(400, "parse_error", result.error)
+        req = result.request
+        if result.skip or req is None:
+            return _error(400, "unparseable", "passthrough not routable "
+                          "without an upstream")
+        if req.prompt_tokens is None and not req.prompt and req.messages:
+            pass  # token-producer will tokenize messages
+        t0 = time.time()
+        handle = service.submit(req)
+        if req.streaming and not req.is_embedding:
+            return StreamingResponse(_sse_stream(req, handle, t0),
+                                     media_type="text/event-stream")
+        completion = await asyncio.to_thread(handle.wait, 120.0)
+        if completion is None:
+            return _error(504, "timeout")
+        if completion.error:
+            status = 429 if completion.error in (
+                "saturated", "queue_capacity", "queue_timeout",
+                "evicted") else 503
+            return _error(status, completion.error)
+        if req.is_embedding:
+            return JSONResponse(_embedding_response(req, completion))
+        return JSONResponse(_completion_response(req, completion, path))
+
+    def _completion_response(req, completion, path):
+        text = tok.decode(completion.tokens)
+        usage = completion.usage.to_openai()
+        created = int(time.time())
+        if "chat" in path:
+            return {"id": f"chatcmpl-{req.request_id}", "object":
+                    "chat.completion", "created": created,
+                    "model": req.model,
+                    "choices": [{"index": 0, "message":
+                                 {"role": "assistant", "content": text},
+                                 "finish_reason": "length"}],
+                    "usage": usage}
+        return {"id": f"cmpl-{req.request_id}", "object": "text_completion",
+                "created": created, "model": req.model,
+                "choices": [{"index": 0, "text": text,
+                             "finish_reason": "length"}],
+                "usage": usage}
+
+    def _embedding_response(req, completion):
+        return {"object": "list", "model": req.model,
+                "data": [{"object": "embedding", "index": 0,
+                          "embedding": completion.tokens or []}],
+                "usage": completion.usage.to_openai()}
+
+    async def _sse_stream(req, handle, t0):
+        idx = 0
+        while True:
+            item = await asyncio.to_thread(handle.token_queue.get)
+            if item is None:
+                break
+            chunk = {"id": f"cmpl-{req.request_id}", "object":
+                     "text_completion.chunk", "model": req.model,
+                     "choices": [{"index": 0, "text": tok.decode([item]),
+                                  "finish_reason": None}]}
+            idx += 1
+            yield f"data: {json.dumps(chunk)}\n\n"
+        completion = handle.completion
+        if completion is not None and not completion.error:
+            final = {"id": f"cmpl-{req.request_id}",
+                     "object": "text_completion.chunk",
+                     "model": req.model,
+                     "choices": [{"index": 0, "text": "",
+                                  "finish_reason": "length"}],
+                     "usage": completion.usage.to_openai()}
+            yield f"data: {json.dumps(final)}\n\n"
+        yield "data: [DONE]\n\n"
+
+    @app.post("/v1/completions")
+    async def completions(request: Request):
+        return await _handle(request, "/v1/completions")
+
+    @app.post("/v1/chat/completions")
+    async def chat_completions(request: Request):
+        return await _handle(request, "/v1/chat/completions")
+
+    @app.post("/v1/responses")
+    async def responses(request: Request):
+        return await _handle(request, "/v1/responses")
+
+    @app.post("/v1/embeddings")
+    async def embeddings(request: Request):
+        return await _handle(request, "/v1/embeddings")
+
+    @app.get("/v1/models")
+    async def models():
+        return {"object": "list", "data": [
+            {"id": service.node.cfg.model.name, "object": "model",
+             "owned_by": "llm-d-inference-scheduler-amd"}]}
+
+    @app.get("/metrics")
+    async def metrics():
+        return Response(content=prom.render(),
+                        media_type="text/plain; version=0.0.4")
+
+    @app.get("/healthz")
+    async def healthz():
+        ready = service.node.datastore.pool_ready()
+        return JSONResponse(status_code=200 if ready else 503,
+                            content={"ready": ready})
+
+    return app

@@ -4,6 +4,8 @@ import numpy as np
 import pytest
 import torch
 
+from llm_d_inference_scheduler_amd.ops.prefix import _i64
+
 pytestmark = pytest.mark.gpu
 
 
@@ -30,7 +32,7 @@ class TestPrefixKernelsGPU:
         seed0 = rc.model_seed("llama-3-8b", "")
         hashes, counts = ext.hash_prompts(
             torch.from_numpy(flat).cuda(),
-            torch.from_numpy(offsets).cuda(), 16, 256, seed0)
+            torch.from_numpy(offsets).cuda(), 16, 256, _i64(seed0))
         hashes = hashes.cpu().numpy()
         counts = counts.cpu().numpy()
         for i, t in enumerate(toks):
@@ -208,3 +210,33 @@ class TestEngineGPU:
             gens.append(toks)
         assert len(gens[0]) == 6
         assert gens[0] == gens[1]
+
+
+class TestGpuPrefixIndex:
+    def test_batched_router_path_matches_host(self):
+        """GpuPrefixIndex batch hash+match == host C++ index for the same
+        insert history (the production router path on GPU boxes)."""
+        from llm_d_inference_scheduler_amd import _router_core as rc
+        from llm_d_inference_scheduler_amd.ops.prefix import GpuPrefixIndex
+        rng = np.random.default_rng(7)
+        gpu = GpuPrefixIndex("cuda:0", capacity_pow2=1 << 16)
+        host = rc.PrefixIndex(10000)
+        seed0 = rc.model_seed("llama-3-8b", "")
+        prompts = [rng.integers(256, 100000, size=int(n)).astype(np.int32)
+                   for n in rng.integers(16, 800, size=12)]
+        # seed both indexes with a few routed prompts
+        for e, p in enumerate(prompts[:6]):
+            h = rc.hash_tokens(p, 16, 256, seed0)
+            host.add(e % 4, h)
+            gpu.add(e % 4, h)
+        # query batch: mix of seen/unseen with shared prefixes
+        queries = [prompts[0], prompts[3][:64],
+                   np.concatenate([prompts[1][:32], prompts[9][:32]]),
+                   prompts[11]]
+        hashes, counts = gpu.hash_prompts_batch(
+            [q.tolist() for q in queries], 16, 256, seed0)
+        match = gpu.match_batch(hashes, counts, 4).cpu().numpy()
+        for i, q in enumerate(queries):
+            hq = rc.hash_tokens(q, 16, 256, seed0)
+            expect = host.match_longest(hq, 4)
+            assert list(match[i]) == list(expect), i

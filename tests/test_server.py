@@ -1,0 +1,91 @@
+"""OpenAI front-door server over the in-process node (tiny CPU config)."""
+import json
+
+import pytest
+import torch
+from fastapi.testclient import TestClient
+
+from llm_d_inference_scheduler_amd.models.configs import TINY_LLAMA
+from llm_d_inference_scheduler_amd.node import NodeConfig, NodeRunner
+from llm_d_inference_scheduler_amd.server import NodeService, build_app
+
+
+@pytest.fixture(scope="module")
+def client():
+    cfg = NodeConfig(model=TINY_LLAMA, world_size=1, topology="mono",
+                     device="cpu", dtype=torch.float32, kv_blocks=256)
+    node = NodeRunner(cfg)
+    service = NodeService(node)
+    service.start()
+    app = build_app(service)
+    with TestClient(app) as c:
+        yield c
+    service.stop()
+
+
+class TestOpenAIServer:
+    def test_completions(self, client):
+        r = client.post("/v1/completions", json={
+            "model": "tiny-llama", "prompt": "hello world test prompt",
+            "max_tokens": 4})
+        assert r.status_code == 200, r.text
+        body = r.json()
+        assert body["object"] == "text_completion"
+        assert body["usage"]["completion_tokens"] == 4
+        assert body["choices"][0]["text"]
+
+    def test_chat_completions(self, client):
+        r = client.post("/v1/chat/completions", json={
+            "model": "tiny-llama",
+            "messages": [{"role": "user", "content": "hi there"}],
+            "max_tokens": 3})
+        assert r.status_code == 200, r.text
+        body = r.json()
+        assert body["choices"][0]["message"]["role"] == "assistant"
+        assert body["usage"]["completion_tokens"] == 3
+
+    def test_streaming_sse(self, client):
+        with client.stream("POST", "/v1/completions", json={
+                "model": "tiny-llama", "prompt": "stream me a response",
+                "max_tokens": 4, "stream": True}) as r:
+            assert r.status_code == 200
+            events = []
+            for line in r.iter_lines():
+                if line.startswith("data:"):
+                    events.append(line[5:].strip())
+        assert events[-1] == "[DONE]"
+        chunks = [json.loads(e) for e in events[:-1]]
+        text_chunks = [c for c in chunks
+                       if c["choices"][0]["text"]]
+        assert len(text_chunks) == 4
+        assert chunks[-1].get("usage", {}).get("completion_tokens") == 4
+
+    def test_embeddings(self, client):
+        r = client.post("/v1/embeddings", json={
+            "model": "tiny-llama", "input": "embed this text"})
+        assert r.status_code == 200, r.text
+        assert r.json()["data"][0]["object"] == "embedding"
+
+    def test_parse_error(self, client):
+        r = client.post("/v1/completions", content=b"{broken",
+                        headers={"content-type": "application/json"})
+        assert r.status_code == 400
+        assert r.headers["x-request-dropped-reason"] == "parse_error"
+
+    def test_missing_model(self, client):
+        r = client.post("/v1/completions", json={"prompt": "x"})
+        assert r.status_code == 400
+
+    def test_metrics_endpoint(self, client):
+        r = client.get("/metrics")
+        assert r.status_code == 200
+        assert b"inference_extension_request_total" in r.content
+        assert b"llm_d_inference_scheduler_disagg_decision_total" in r.content
+
+    def test_healthz(self, client):
+        r = client.get("/healthz")
+        assert r.status_code == 200 and r.json()["ready"]
+
+    def test_models(self, client):
+        r = client.get("/v1/models")
+        assert r.json()["data"][0]["id"] == "tiny-llama"

@@ -60,6 +60,51 @@ def pd_topology(n: int) -> str:
     return f"pd:{prefill}p{n - prefill}d"
 
 
+def epd_topology(n: int) -> str:
+    # BASELINE config 5: 1 encode + 2 prefill + 5 decode at n=8
+    if n < 3:
+        return pd_topology(n)
+    prefill = max(1, (n - 1) // 3)
+    return f"epd:1e{prefill}p{n - 1 - prefill}d"
+
+
+EPD_YAML = """
+plugins:
+  - type: decode-filter
+  - type: prefill-filter
+  - type: encode-filter
+  - type: queue-scorer
+  - type: kv-cache-utilization-scorer
+  - type: prefix-cache-scorer
+  - type: max-score-picker
+  - type: prefix-based-pd-decider
+    parameters: {nonCachedTokens: 256}
+  - type: always-disagg-multimodal-decider
+  - type: disagg-profile-handler
+    parameters:
+      pdDecider: prefix-based-pd-decider
+      encodeDecider: always-disagg-multimodal-decider
+schedulingProfiles:
+  - name: decode
+    plugins:
+      - {pluginRef: decode-filter}
+      - {pluginRef: prefix-cache-scorer, weight: 3}
+      - {pluginRef: queue-scorer, weight: 1}
+      - {pluginRef: kv-cache-utilization-scorer, weight: 1}
+      - {pluginRef: max-score-picker}
+  - name: prefill
+    plugins:
+      - {pluginRef: prefill-filter}
+      - {pluginRef: queue-scorer, weight: 1}
+      - {pluginRef: max-score-picker}
+  - name: encode
+    plugins:
+      - {pluginRef: encode-filter}
+      - {pluginRef: queue-scorer, weight: 1}
+      - {pluginRef: max-score-picker}
+"""
+
+
 class Workload:
     """Deterministic shared-prefix synthetic prompt stream."""
 
@@ -86,6 +131,12 @@ class Workload:
                          prompt_tokens=tokens, max_tokens=a.max_tokens,
                          prompt="")
         req.ttft_slo_ms = a.ttft_slo_ms
+        if a.mode == "epd":
+            from llm_d_inference_scheduler_amd.scheduling.types import \
+                MultiModalItem
+            # 1 image per request from a small URL pool (dedupe-friendly)
+            req.mm_items = [MultiModalItem("image_url",
+                                           f"http://img/{i % 16}")]
         if a.mode == "fc":
             # mixed-SLO priority tiers: 1/3 critical, 1/3 standard,
             # 1/3 sheddable batch (InferenceObjective priorities)
@@ -120,32 +171,40 @@ def main():
         transfer_group = None  # default group (RCCL over xGMI)
         mailbox_group = dist.new_group(backend="gloo")
 
-    from llm_d_inference_scheduler_amd.models.configs import (LLAMA_3_8B,
-                                                              TINY_LLAMA)
+    from llm_d_inference_scheduler_amd.models.configs import (
+        LLAMA_3_8B, LLAVA_1_5_7B_TEXT, TINY_LLAMA, TINY_LLAVA)
     from llm_d_inference_scheduler_amd.node import NodeConfig, NodeRunner
     from llm_d_inference_scheduler_amd.flowcontrol import BandConfig
     from llm_d_inference_scheduler_amd.api.objectives import InferenceObjective
 
     model_cfg = LLAMA_3_8B if args.model == "llama-3-8b" else TINY_LLAMA
-    if not use_gpu and args.model == "llama-3-8b":
+    if args.mode == "epd":
+        model_cfg = LLAVA_1_5_7B_TEXT  # BASELINE config 5 multimodal model
+        args.model = model_cfg.name
+    if not use_gpu and args.model in ("llama-3-8b", "llava-1.5-7b"):
         # CPU smoke of the bench harness itself uses the tiny config; a GPU
         # run always uses the full flagship model (anything else is invalid
         # for reporting).
-        model_cfg = TINY_LLAMA
+        model_cfg = TINY_LLAVA if args.mode == "epd" else TINY_LLAMA
         args.prompt_len = min(args.prompt_len, 96)
         args.max_tokens = min(args.max_tokens, 8)
         args.concurrency = min(args.concurrency, 8)
 
     topology = "mono"
     parallelism = f"dp{world}"
-    if args.mode in ("pd", "epd") and world >= 2:
+    epp_yaml = ""
+    if args.mode == "pd" and world >= 2:
         topology = pd_topology(world)
         parallelism = topology
+    elif args.mode == "epd":
+        topology = epd_topology(world)
+        parallelism = topology
+        epp_yaml = EPD_YAML
     fc = args.mode == "fc"
 
     cfg = NodeConfig(
         model=model_cfg, rank=rank, world_size=world, topology=topology,
-        device=device, dtype=dtype,
+        epp_yaml=epp_yaml, device=device, dtype=dtype,
         kv_blocks=None if use_gpu else 2048,
         kv_budget_bytes=int(args.kv_gb * (1 << 30)),
         flow_control=fc,

@@ -30,6 +30,9 @@ class EngineRequest:
     is_embedding: bool = False
     prefill_only: bool = False      # disagg: stop after prefill + 1st token
     cached_tokens: int = 0          # router-estimated prefix reuse (metrics)
+    # multimodal: embeddings for the first len(prefix_embeds) prompt rows
+    # (prompt_tokens must carry placeholder ids for those positions)
+    prefix_embeds: "Optional[torch.Tensor]" = None
     # runtime state
     computed: int = 0               # prompt tokens already prefilled
     generated: List[int] = field(default_factory=list)
@@ -188,9 +191,15 @@ class EngineWorker:
 
         input_ids, positions, slots = [], [], []
         seq_starts, ctx_lens, tables, logit_rows = [0], [], [], []
+        embed_rows, embed_vals = [], []
         finishing: List[EngineRequest] = []
         for req, chunk in selected:
             start, end = req.computed, req.computed + chunk
+            if req.prefix_embeds is not None and start < req.prefix_embeds.shape[0]:
+                e_end = min(end, req.prefix_embeds.shape[0])
+                base_row = seq_starts[-1]
+                embed_rows.extend(range(base_row, base_row + (e_end - start)))
+                embed_vals.append(req.prefix_embeds[start:e_end])
             input_ids.extend(req.prompt_tokens[start:end])
             positions.extend(range(start, end))
             slots.extend(self.mgr.slots_for_range(req.request_id, start, end))
@@ -217,7 +226,12 @@ class EngineWorker:
             logit_rows=torch.tensor(logit_rows, dtype=torch.int64,
                                     device=self.device)
             if logit_rows else torch.zeros(0, dtype=torch.int64,
-                                           device=self.device))
+                                           device=self.device),
+            embed_rows=torch.tensor(embed_rows, dtype=torch.int64,
+                                    device=self.device)
+            if embed_rows else None,
+            embed_values=torch.cat(embed_vals).to(self.device)
+            if embed_vals else None)
         embedding_reqs = [r for r in finishing if r.is_embedding]
         hidden_or_logits = self.model.forward(
             batch, self.pool.tensor, embeddings_out=bool(embedding_reqs))

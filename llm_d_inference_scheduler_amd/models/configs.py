@@ -49,3 +49,9 @@ TINY_LLAMA = ModelConfig(
     name="tiny-llama", vocab_size=1024, hidden_size=256,
     intermediate_size=512, num_layers=2, num_heads=8, num_kv_heads=4,
     head_dim=32, rope_theta=10000.0, max_position=512)
+
+TINY_LLAVA = ModelConfig(
+    name="tiny-llava", vocab_size=1024, hidden_size=256,
+    intermediate_size=512, num_layers=2, num_heads=8, num_kv_heads=4,
+    head_dim=32, rope_theta=10000.0, max_position=512,
+    vision_hidden=64, vision_layers=2, vision_patches=8)

@@ -41,6 +41,9 @@ class ForwardBatch:
     ctx_lens: Optional[List[int]] = None          # [B] total ctx after chunk
     prefill_block_tables: Optional[List[torch.Tensor]] = None
     logit_rows: Optional[torch.Tensor] = None     # rows needing logits
+    # multimodal: rows whose input embedding is provided (encode hand-off)
+    embed_rows: Optional[torch.Tensor] = None     # [n] int64 into T
+    embed_values: Optional[torch.Tensor] = None   # [n, hidden]
 
 
 class LlamaRunner:
@@ -104,6 +107,8 @@ class LlamaRunner:
         embeddings_out). kv_pool: [L, 2, NB, KVH, BS, D]."""
         c = self.cfg
         hidden = self.embed[batch.input_ids]
+        if batch.embed_rows is not None and batch.embed_rows.numel():
+            hidden[batch.embed_rows] = batch.embed_values.to(hidden.dtype)
         residual = None
         pos32 = batch.positions.to(torch.int32)
         for li, layer in enumerate(self.layers):

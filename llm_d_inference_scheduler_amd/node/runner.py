@@ -116,6 +116,15 @@ class NodeRunner:
             dtype=cfg.dtype, prefill_chunk_tokens=cfg.prefill_chunk_tokens,
             max_decode_batch=cfg.max_decode_batch,
             ttft_slo_ms=cfg.ttft_slo_ms, seed=cfg.seed)
+        # encode role: vision tower + URL-deduped embedding cache
+        self.encoder = None
+        if (self.my_spec.role & Role.ENCODE) and cfg.model.vision_hidden:
+            from ..models.vision import VisionEncoder
+            self.encoder = VisionEncoder(cfg.model, cfg.device,
+                                         dtype=cfg.dtype, seed=cfg.seed)
+        self._encode_jobs: List[Dict[str, Any]] = []
+        self._emb_pending: Dict[str, torch.Tensor] = {}  # src side
+        self._awaiting_embeds: Dict[str, Dict[str, Any]] = {}  # dst side
         self.transfer = KVTransferEngine(self.engine.pool.tensor, self.rank,
                                          group=cfg.transfer_group)
         self._outbox: List[Dict[str, Any]] = []
@@ -230,6 +239,7 @@ class NodeRunner:
         self._outbox = []
         self._process_messages(msgs)
         self._execute_transfers(msgs)
+        self._run_encoder()
         outputs = self.engine.step()
         self._handle_outputs(outputs)
         self._step += 1
@@ -331,9 +341,16 @@ class NodeRunner:
             if prefill_hdr:
                 msg["prefill"] = int(prefill_hdr.split(":")[-1])
             encode_hdr = req.headers.get("x-encoder-hosts-ports")
-            if encode_hdr:
+            if encode_hdr and req.mm_items:
                 msg["encode"] = [int(h.split(":")[-1])
                                  for h in encode_hdr.split(",")]
+                # dedupe by URL (connector_epd_shared_storage.go:125-208)
+                seen, urls = set(), []
+                for item in req.mm_items:
+                    if item.url not in seen:
+                        seen.add(item.url)
+                        urls.append(item.url)
+                msg["mm"] = urls
             self._outbox.append(msg)
 
     def _cached_tokens(self, decision: RoutingDecision,
@@ -376,22 +393,39 @@ class NodeRunner:
     def _handle_assign(self, m: Dict[str, Any]) -> None:
         prefill_rank = m.get("prefill")
         decode_rank = m["dst"]
+        encode_ranks = m.get("encode") or []
         req = EngineRequest(
             request_id=m["req_id"], prompt_tokens=list(m["tokens"]),
             max_tokens=m["max_tokens"], temperature=m["temperature"],
             is_embedding=m.get("is_embedding", False),
             cached_tokens=m.get("cached", 0),
             arrival_t=m.get("arrival") or 0.0)
-        if prefill_rank is not None and prefill_rank != decode_rank:
-            if self.rank == prefill_rank:
+        # the rank that runs the prompt (prefill stage or monolithic decode)
+        prompt_rank = prefill_rank if (
+            prefill_rank is not None and prefill_rank != decode_rank) \
+            else decode_rank
+        if self.rank == prompt_rank:
+            if prompt_rank != decode_rank:
                 req.prefill_only = True
                 self._handoff_dst[m["req_id"]] = decode_rank
+            if encode_ranks:
+                # E stage first: hold until embeddings arrive over xGMI
+                self._awaiting_embeds[m["req_id"]] = {
+                    "req": req, "src": encode_ranks[0],
+                    "n_mm": len(m.get("mm", []))}
+            else:
                 self.engine.add_request(req)
-            elif self.rank == decode_rank:
-                self._pending_adoption[m["req_id"]] = {"req": req,
-                                                       "src": prefill_rank}
-        elif self.rank == decode_rank:
-            self.engine.add_request(req)
+        if self.rank == decode_rank and prompt_rank != decode_rank:
+            self._pending_adoption[m["req_id"]] = {"req": EngineRequest(
+                request_id=m["req_id"], prompt_tokens=list(m["tokens"]),
+                max_tokens=m["max_tokens"], temperature=m["temperature"],
+                cached_tokens=m.get("cached", 0),
+                arrival_t=m.get("arrival") or 0.0),
+                "src": prompt_rank}
+        if encode_ranks and self.rank in encode_ranks:
+            self._encode_jobs.append({"req_id": m["req_id"],
+                                      "urls": m.get("mm", []),
+                                      "dst": prompt_rank})
         if m.get("stream") and self.rank == decode_rank:
             self._streaming_ids.add(m["req_id"])
 
@@ -408,8 +442,48 @@ class NodeRunner:
             request_id=m["req_id"], usage=usage,
             tokens=m.get("tokens", []), error=m.get("error", "")))
 
+    # ---- encode stage (E/PD, E/P/D) ----
+    def _run_encoder(self, max_jobs: int = 8) -> None:
+        if self.encoder is None or not self._encode_jobs:
+            return
+        jobs, self._encode_jobs = (self._encode_jobs[:max_jobs],
+                                   self._encode_jobs[max_jobs:])
+        for job in jobs:
+            embs = [self.encoder.encode_url(u) for u in job["urls"]]
+            emb = torch.cat(embs) if embs else torch.zeros(
+                (0, self.cfg.model.hidden_size), dtype=self.cfg.dtype)
+            if job["dst"] == self.rank:
+                self._attach_embeds(job["req_id"], emb)
+                continue
+            self._emb_pending[job["req_id"]] = emb
+            self._outbox.append({"type": "emb_ready",
+                                 "req_id": job["req_id"], "src": self.rank,
+                                 "dst": job["dst"],
+                                 "rows": int(emb.shape[0])})
+
+    def _attach_embeds(self, req_id: str, emb: torch.Tensor) -> None:
+        pending = self._awaiting_embeds.pop(req_id, None)
+        if pending is None:
+            return
+        req = pending["req"]
+        req.prefix_embeds = emb
+        req.prompt_tokens = [0] * emb.shape[0] + req.prompt_tokens
+        self.engine.add_request(req)
+
     # ---- transfers (the NIXL-v2 step 2/3 replacement) ----
     def _execute_transfers(self, msgs: List[Dict[str, Any]]) -> None:
+        emb_jobs = sorted((m for m in msgs if m.get("type") == "emb_ready"),
+                          key=lambda m: m["req_id"])
+        for job in emb_jobs:
+            src, dst = job["src"], job["dst"]
+            if self.rank == src:
+                self.transfer.send_tensor(dst,
+                                          self._emb_pending.pop(job["req_id"]))
+            elif self.rank == dst:
+                emb = self.transfer.recv_tensor(
+                    src, (job["rows"], self.cfg.model.hidden_size),
+                    self.cfg.dtype)
+                self._attach_embeds(job["req_id"], emb)
         jobs = sorted((m for m in msgs if m.get("type") == "kv_ready"),
                       key=lambda m: m["req_id"])
         for job in jobs:

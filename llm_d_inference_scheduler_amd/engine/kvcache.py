@@ -1,18 +1,39 @@
-"""Paged KV pool + block allocator (per-GPU worker engine).
+"""Paged KV pool + prefix-caching block allocator (per-GPU worker engine).
 
 Pool layout [L, 2, NB, KVH, BS, D] bf16 — one (layer, K/V, block) is a
 contiguous MFMA-tile-aligned chunk consumed zero-repack by the decode
 attention kernel and the xGMI transfer engine (kv_cache.hip
 header). Sizing targets the 288 GB HBM3E budget: Llama-3-8B bf16 KV is
 128 KB/token -> ~1.9M tokens of residency beside the 16 GB of weights.
-"""
-from typing import Dict, List, Optional
 
+The BlockManager implements engine-level automatic prefix caching: full
+blocks are content-addressed by the same chained xxhash64 the router's
+approx-prefix producer uses (reference approximateprefix/hashing.go:35-99 —
+there it only *estimates* vLLM's cache; here the engine IS the model server,
+so the cache is real). Freed blocks keep their content in an LRU and are
+resurrected on hash hit; eviction happens only on reuse.
+"""
+from collections import OrderedDict
+from typing import Dict, List, Optional, Sequence
+
+import numpy as np
 import torch
 
 from ..models.configs import ModelConfig
 
 BLOCK_SIZE = 16  # tokens per KV block (reference default, types.go:92)
+
+
+def block_hashes(prompt_tokens: Sequence[int], block_size: int = BLOCK_SIZE,
+                 seed: int = 0x9E3779B97F4A7C15) -> np.ndarray:
+    """Chained per-block content hashes of the full prompt blocks."""
+    from .. import _router_core as rc
+    toks = np.asarray(prompt_tokens, dtype=np.int32)
+    n_full = len(toks) // block_size
+    if n_full == 0:
+        return np.zeros(0, dtype=np.uint64)
+    return rc.hash_tokens(toks[:n_full * block_size], block_size, n_full,
+                          seed)
 
 
 class KVPool:
@@ -48,36 +69,111 @@ class KVPool:
 
 
 class BlockManager:
-    """Free-list allocator + per-sequence block tables."""
+    """Refcounted free-list allocator + per-sequence block tables with
+    content-addressed full-block prefix caching."""
 
-    def __init__(self, num_blocks: int, block_size: int = BLOCK_SIZE):
+    def __init__(self, num_blocks: int, block_size: int = BLOCK_SIZE,
+                 prefix_caching: bool = True):
         self.num_blocks = num_blocks
         self.block_size = block_size
-        self._free: List[int] = list(range(num_blocks - 1, -1, -1))
+        self.prefix_caching = prefix_caching
+        # all blocks start free and content-less; LRU order: oldest first
+        self._free_lru: "OrderedDict[int, None]" = OrderedDict(
+            (b, None) for b in range(num_blocks))
+        self._refcnt = [0] * num_blocks
+        self._hash_of: List[Optional[int]] = [None] * num_blocks
+        self._by_hash: Dict[int, int] = {}
         self.tables: Dict[str, List[int]] = {}
         self.seq_lens: Dict[str, int] = {}
+        # stats
+        self.cached_tokens_total = 0
+        self.queried_tokens_total = 0
 
     @property
     def free_blocks(self) -> int:
-        return len(self._free)
+        return len(self._free_lru)
 
     @property
     def usage(self) -> float:
-        return 1.0 - len(self._free) / max(1, self.num_blocks)
+        return 1.0 - len(self._free_lru) / max(1, self.num_blocks)
 
+    @property
+    def hit_rate(self) -> float:
+        q = self.queried_tokens_total
+        return self.cached_tokens_total / q if q else 0.0
+
+    # ---- free-list internals -------------------------------------------
+    def _pop_free(self) -> int:
+        blk, _ = self._free_lru.popitem(last=False)
+        h = self._hash_of[blk]
+        if h is not None:                      # evict stale cached content
+            if self._by_hash.get(h) == blk:
+                del self._by_hash[h]
+            self._hash_of[blk] = None
+        self._refcnt[blk] = 1
+        return blk
+
+    def _release(self, blk: int) -> None:
+        self._refcnt[blk] -= 1
+        if self._refcnt[blk] == 0:
+            # most-recently-freed = most likely reused: push to LRU tail
+            self._free_lru[blk] = None
+
+    # ---- prefix caching -------------------------------------------------
+    def match_prefix(self, hashes: np.ndarray, prompt_len: int) -> int:
+        """Longest cached full-block prefix, capped so at least one prompt
+        token is always recomputed (its logits seed generation)."""
+        if not self.prefix_caching:
+            return 0
+        max_blocks = min(len(hashes), (prompt_len - 1) // self.block_size)
+        n = 0
+        while n < max_blocks and int(hashes[n]) in self._by_hash:
+            n += 1
+        return n * self.block_size
+
+    def allocate_prompt(self, seq_id: str, hashes: np.ndarray,
+                        prompt_len: int) -> int:
+        """Build the sequence's table reusing cached prefix blocks.
+        Returns the number of cached (skippable) prompt tokens."""
+        matched = self.match_prefix(hashes, prompt_len)
+        table = self.tables.setdefault(seq_id, [])
+        assert not table, "allocate_prompt on an existing sequence"
+        for i in range(matched // self.block_size):
+            blk = self._by_hash[int(hashes[i])]
+            if self._refcnt[blk] == 0:          # resurrect from free LRU
+                del self._free_lru[blk]
+            self._refcnt[blk] += 1
+            table.append(blk)
+        self.seq_lens[seq_id] = 0
+        self.queried_tokens_total += prompt_len
+        self.cached_tokens_total += matched
+        return matched
+
+    def register_block(self, seq_id: str, block_idx: int, h: int) -> None:
+        """Publish a fully-written block's content hash (prefill only —
+        shared blocks are immutable: writers always own fresh blocks)."""
+        if not self.prefix_caching:
+            return
+        blk = self.tables[seq_id][block_idx]
+        if h in self._by_hash or self._hash_of[blk] is not None:
+            return                              # first writer wins
+        self._by_hash[h] = blk
+        self._hash_of[blk] = h
+
+    # ---- allocation ------------------------------------------------------
     def can_allocate(self, n_tokens: int) -> bool:
         need = (n_tokens + self.block_size - 1) // self.block_size
-        return need <= len(self._free)
+        return need <= len(self._free_lru)
 
     def allocate(self, seq_id: str, n_tokens: int) -> bool:
         """Ensure the sequence has capacity for n_tokens total."""
         table = self.tables.setdefault(seq_id, [])
         need = (n_tokens + self.block_size - 1) // self.block_size
+        if need - len(table) > len(self._free_lru):
+            return False
         while len(table) < need:
-            if not self._free:
-                return False
-            table.append(self._free.pop())
-        self.seq_lens[seq_id] = max(self.seq_lens.get(seq_id, 0), 0)
+            table.append(self._pop_free())
+        self.seq_lens.setdefault(seq_id, 0)
         return True
 
     def append_token_slot(self, seq_id: str) -> Optional[int]:
@@ -100,7 +196,7 @@ class BlockManager:
 
     def free(self, seq_id: str) -> None:
         for blk in self.tables.pop(seq_id, []):
-            self._free.append(blk)
+            self._release(blk)
         self.seq_lens.pop(seq_id, None)
 
     def adopt(self, seq_id: str, blocks: List[int], seq_len: int) -> None:
@@ -109,9 +205,10 @@ class BlockManager:
         self.seq_lens[seq_id] = seq_len
 
     def take_blocks(self, n: int) -> Optional[List[int]]:
-        if n > len(self._free):
+        if n > len(self._free_lru):
             return None
-        return [self._free.pop() for _ in range(n)]
+        return [self._pop_free() for _ in range(n)]
 
     def release_blocks(self, blocks: List[int]) -> None:
-        self._free.extend(blocks)
+        for blk in blocks:
+            self._release(blk)

@@ -5,6 +5,14 @@ pods (the router only steers them). One EngineWorker owns
 one GPU: chunked prefill + batched decode (gfx950 paged-attention kernel),
 role-aware behavior (decode | prefill | encode | combinations), and the
 vLLM-compatible metrics snapshot the datalayer collectors scrape.
+
+Decode hot loop design (MI355X-first): all per-step state — block tables,
+sequence lengths, last sampled token ids — is device-resident; a step is
+assembled and launched with no host→device traffic except on block-boundary
+crossings and batch-membership changes, and sampled tokens are read back one
+step late through a pinned buffer + event (the host never blocks the GPU on
+a same-step sync). Finish decisions are made from host-side token *counts*,
+so step N+1 launches before step N's token values have landed.
 """
 import time
 from dataclasses import dataclass, field
@@ -16,7 +24,7 @@ from ..datalayer.endpoint import Metrics, Role
 from ..models.configs import ModelConfig
 from ..models.llama import ForwardBatch, LlamaRunner
 from ..utils.logging import get_logger
-from .kvcache import BLOCK_SIZE, BlockManager, KVPool
+from .kvcache import BLOCK_SIZE, BlockManager, KVPool, block_hashes
 
 log = get_logger("engine.worker")
 
@@ -29,16 +37,19 @@ class EngineRequest:
     temperature: float = 0.0
     is_embedding: bool = False
     prefill_only: bool = False      # disagg: stop after prefill + 1st token
-    cached_tokens: int = 0          # router-estimated prefix reuse (metrics)
+    cached_tokens: int = 0          # prefix-cache reuse (engine-measured)
     # multimodal: embeddings for the first len(prefix_embeds) prompt rows
     # (prompt_tokens must carry placeholder ids for those positions)
     prefix_embeds: "Optional[torch.Tensor]" = None
     # runtime state
     computed: int = 0               # prompt tokens already prefilled
     generated: List[int] = field(default_factory=list)
+    inflight: int = 0               # sampled on device, not yet collected
     slo_ok: bool = True             # TTFT met the SLO (goodput accounting)
     arrival_t: float = 0.0
     first_token_t: float = 0.0
+    block_hashes: Optional[object] = None   # np.uint64 per full prompt block
+    registered_blocks: int = 0              # hashes published so far
 
     def __post_init__(self):
         if not self.arrival_t:
@@ -47,6 +58,11 @@ class EngineRequest:
     @property
     def prompt_len(self) -> int:
         return len(self.prompt_tokens)
+
+    @property
+    def token_count(self) -> int:
+        """Generated + in-flight tokens (host-known without a sync)."""
+        return len(self.generated) + self.inflight
 
 
 @dataclass
@@ -71,6 +87,70 @@ class RequestOutput:
     e2e_ms: Optional[float] = None
 
 
+class DecodeState:
+    """Device-resident decode batch state (grows on demand)."""
+
+    def __init__(self, capacity: int, max_blocks: int, device: torch.device):
+        self.device = device
+        self.max_blocks = max_blocks
+        self.capacity = 0
+        self.bt = None
+        self.seq_lens = None
+        self.last_tok = None
+        self.temps = None
+        self.free_rows: List[int] = []
+        self.row_of: Dict[str, int] = {}
+        self._active_ids: List[str] = []
+        self._rows_t: Optional[torch.Tensor] = None
+        self._grow(capacity)
+
+    def _grow(self, new_cap: int) -> None:
+        old = self.capacity
+        bt = torch.zeros((new_cap, self.max_blocks), dtype=torch.int32,
+                         device=self.device)
+        sl = torch.zeros(new_cap, dtype=torch.int32, device=self.device)
+        lt = torch.zeros(new_cap, dtype=torch.int64, device=self.device)
+        tp = torch.zeros(new_cap, dtype=torch.float32, device=self.device)
+        if old:
+            bt[:old] = self.bt
+            sl[:old] = self.seq_lens
+            lt[:old] = self.last_tok
+            tp[:old] = self.temps
+        self.bt, self.seq_lens, self.last_tok, self.temps = bt, sl, lt, tp
+        self.free_rows.extend(range(new_cap - 1, old - 1, -1))
+        self.capacity = new_cap
+
+    def take_row(self) -> int:
+        if not self.free_rows:
+            self._grow(max(8, self.capacity * 2))
+        return self.free_rows.pop()
+
+    def join(self, req: EngineRequest, table: List[int], seq_len: int,
+             last_token: int) -> None:
+        row = self.take_row()
+        self.row_of[req.request_id] = row
+        if len(table) > self.max_blocks:
+            raise RuntimeError("sequence exceeds decode-state max_blocks")
+        self.bt[row, :len(table)] = torch.tensor(
+            table, dtype=torch.int32).to(self.device, non_blocking=True)
+        self.seq_lens[row] = seq_len
+        self.last_tok[row] = last_token
+        self.temps[row] = req.temperature
+
+    def leave(self, request_id: str) -> None:
+        row = self.row_of.pop(request_id, None)
+        if row is not None:
+            self.free_rows.append(row)
+
+    def rows_for(self, ids: List[str]) -> torch.Tensor:
+        if ids != self._active_ids:
+            self._rows_t = torch.tensor(
+                [self.row_of[i] for i in ids], dtype=torch.int64,
+                device=self.device)
+            self._active_ids = list(ids)
+        return self._rows_t
+
+
 class EngineWorker:
     def __init__(self, config: ModelConfig, device,
                  role: Role = Role.DECODE,
@@ -81,6 +161,7 @@ class EngineWorker:
                  max_decode_batch: int = 256,
                  max_model_len: int = 8192,
                  ttft_slo_ms: float = None,
+                 prefix_caching: bool = True,
                  seed: int = 0):
         self.cfg = config
         self.device = torch.device(device)
@@ -94,11 +175,20 @@ class EngineWorker:
             kv_blocks = KVPool.blocks_for_budget(config, kv_budget_bytes,
                                                  dtype_bytes=2)
         self.pool = KVPool(config, kv_blocks, self.device, dtype)
-        self.mgr = BlockManager(kv_blocks, self.pool.block_size)
+        self.mgr = BlockManager(kv_blocks, self.pool.block_size,
+                                prefix_caching=prefix_caching)
         self.model = LlamaRunner(config, self.device, dtype, seed=seed)
         self.waiting: List[EngineRequest] = []
         self.running: List[EngineRequest] = []
         self._by_id: Dict[str, EngineRequest] = {}
+        # rows must hold prompt + generation; generation length is
+        # unbounded by max_model_len in this engine, so budget 2x
+        max_blocks = (2 * max_model_len) // self.pool.block_size + 2
+        self.dstate = DecodeState(min(64, max_decode_batch), max_blocks,
+                                  self.device)
+        self._cuda = self.device.type == "cuda"
+        self._pin = None
+        self._pending = None   # (reqs, event|None, n) — one-step readback lag
         self.steps = 0
         self.total_generated = 0
         self.total_generated_slo = 0
@@ -130,6 +220,7 @@ class EngineWorker:
             req.slo_ok = False
         self._by_id[req.request_id] = req
         self.running.append(req)
+        self.dstate.join(req, local_blocks, seq_len, first_token)
 
     def abort(self, request_id: str) -> None:
         req = self._by_id.pop(request_id, None)
@@ -139,6 +230,7 @@ class EngineWorker:
             self.waiting.remove(req)
         if req in self.running:
             self.running.remove(req)
+        self.dstate.leave(request_id)
         self.mgr.free(request_id)
 
     # ------------------------------------------------------------------
@@ -153,27 +245,44 @@ class EngineWorker:
 
     @property
     def has_work(self) -> bool:
-        return bool(self.waiting or self.running)
+        return bool(self.waiting or self.running or self._pending)
 
     # ------------------------------------------------------------------
     def step(self) -> List[RequestOutput]:
         """One engine iteration: a chunked-prefill pass (if any waiting) and
-        one decode pass over the running batch."""
+        one decode pass over the running batch. Decode token values surface
+        one step late (pipelined readback)."""
         outputs: List[RequestOutput] = []
         if self.waiting:
             outputs.extend(self._prefill_pass())
         if self.running:
             outputs.extend(self._decode_pass())
+        elif self._pending is not None:
+            outputs.extend(self._collect_pending())
         self.steps += 1
         return outputs
 
     # ---- prefill ----
+    def _admit_prompt(self, req: EngineRequest) -> None:
+        """First touch: match + reuse cached prefix blocks (engine APC)."""
+        if req.block_hashes is None:
+            req.block_hashes = block_hashes(req.prompt_tokens,
+                                            self.pool.block_size)
+        matched = self.mgr.allocate_prompt(req.request_id, req.block_hashes,
+                                           req.prompt_len)
+        req.computed = matched
+        req.cached_tokens = matched
+        req.registered_blocks = matched // self.pool.block_size
+        self.mgr.set_seq_len(req.request_id, matched)
+
     def _prefill_pass(self) -> List[RequestOutput]:
         budget = self.prefill_chunk_tokens
         selected: List[tuple] = []
         for req in list(self.waiting):
             if budget <= 0:
                 break
+            if req.request_id not in self.mgr.tables:
+                self._admit_prompt(req)
             remaining = req.prompt_len - req.computed
             if remaining <= 0:
                 self.waiting.remove(req)
@@ -193,11 +302,12 @@ class EngineWorker:
         seq_starts, ctx_lens, tables, logit_rows = [0], [], [], []
         embed_rows, embed_vals = [], []
         finishing: List[EngineRequest] = []
+        bs = self.pool.block_size
         for req, chunk in selected:
             start, end = req.computed, req.computed + chunk
             if req.prefix_embeds is not None and start < req.prefix_embeds.shape[0]:
                 e_end = min(end, req.prefix_embeds.shape[0])
-                base_row = seq_starts[-1]
+                base_row = seq_starts[-1] + 0
                 embed_rows.extend(range(base_row, base_row + (e_end - start)))
                 embed_vals.append(req.prefix_embeds[start:e_end])
             input_ids.extend(req.prompt_tokens[start:end])
@@ -213,6 +323,13 @@ class EngineWorker:
                 finishing.append(req)
             req.computed = end
             self.total_prefilled += chunk
+            # publish content hashes of now-complete blocks (prefill only;
+            # shared cached blocks are immutable by construction)
+            full = end // bs
+            for b in range(req.registered_blocks, full):
+                self.mgr.register_block(req.request_id, b,
+                                        int(req.block_hashes[b]))
+            req.registered_blocks = max(req.registered_blocks, full)
 
         batch = ForwardBatch(
             input_ids=torch.tensor(input_ids, dtype=torch.int64,
@@ -267,7 +384,7 @@ class EngineWorker:
         return outputs
 
     def _finish_prefills(self, finishing, logits, outputs) -> None:
-        tokens = self._sample(logits, [r.temperature for r in finishing])
+        tokens = self._sample_host(logits, [r.temperature for r in finishing])
         now = time.time()
         for req, tok in zip(finishing, tokens):
             req.first_token_t = now
@@ -291,6 +408,8 @@ class EngineWorker:
                 if req.slo_ok:
                     self.total_generated_slo += 1
                 self.running.append(req)
+                self.dstate.join(req, self.mgr.tables[req.request_id],
+                                 self.mgr.seq_lens[req.request_id], int(tok))
                 outputs.append(RequestOutput(
                     request_id=req.request_id, new_tokens=[int(tok)],
                     ttft_ms=(now - req.arrival_t) * 1e3,
@@ -303,41 +422,93 @@ class EngineWorker:
 
     # ---- decode ----
     def _decode_pass(self) -> List[RequestOutput]:
-        batch_reqs = self.running[:self.max_decode_batch]
-        input_ids, positions, slots, seq_lens, tables = [], [], [], [], []
+        """Assemble and launch step N, then collect step N-1's tokens.
+        Membership and finishes are decided from host-side counts, so the
+        launch never waits on token values."""
+        st = self.dstate
+        bs = self.pool.block_size
         active: List[EngineRequest] = []
-        for req in batch_reqs:
-            slot = self.mgr.append_token_slot(req.request_id)
-            if slot is None:
-                continue  # out of KV blocks: stall this sequence
+        upd_rows: List[int] = []
+        upd_idx: List[int] = []
+        upd_val: List[int] = []
+        done_now: List[EngineRequest] = []
+        for req in self.running:
+            if len(active) >= self.max_decode_batch:
+                break
+            if req.token_count >= req.max_tokens:
+                if req.inflight == 0:
+                    done_now.append(req)  # all tokens came from prefill
+                continue          # else final token in flight; collect soon
+            rid = req.request_id
+            cur = self.mgr.seq_lens[rid]
+            if cur % bs == 0 and len(self.mgr.tables[rid]) <= cur // bs:
+                if not self.mgr.allocate(rid, cur + 1):
+                    continue      # out of KV blocks: stall this sequence
+                upd_rows.append(st.row_of[rid])
+                upd_idx.append(cur // bs)
+                upd_val.append(self.mgr.tables[rid][-1])
+            self.mgr.set_seq_len(rid, cur + 1)
+            req.inflight += 1
             active.append(req)
-            input_ids.append(req.generated[-1])
-            positions.append(self.mgr.seq_lens[req.request_id] - 1)
-            slots.append(slot)
-            seq_lens.append(self.mgr.seq_lens[req.request_id])
-            tables.append(self.mgr.tables[req.request_id])
+
+        finals = [self._finalize(r) for r in done_now]
         if not active:
-            return []
-        max_blocks = max(len(t) for t in tables)
-        bt = torch.zeros((len(active), max_blocks), dtype=torch.int32)
-        for i, t in enumerate(tables):
-            bt[i, :len(t)] = torch.tensor(t, dtype=torch.int32)
+            return self._collect_pending() + finals
+
+        if upd_rows:
+            st.bt[torch.tensor(upd_rows, dtype=torch.int64, device=self.device),
+                  torch.tensor(upd_idx, dtype=torch.int64, device=self.device)
+                  ] = torch.tensor(upd_val, dtype=torch.int32,
+                                   device=self.device)
+
+        rows_t = st.rows_for([r.request_id for r in active])
+        lens = st.seq_lens.index_select(0, rows_t)
+        new_len = lens + 1
+        pos = new_len - 1                           # int32 positions
+        bt_rows = st.bt.index_select(0, rows_t)
+        blk = bt_rows.gather(
+            1, (pos // bs).to(torch.int64).unsqueeze(1)).squeeze(1)
+        slot = blk.to(torch.int64) * bs + (pos % bs).to(torch.int64)
+        st.seq_lens.index_copy_(0, rows_t, new_len)
         batch = ForwardBatch(
-            input_ids=torch.tensor(input_ids, dtype=torch.int64,
-                                   device=self.device),
-            positions=torch.tensor(positions, dtype=torch.int32,
-                                   device=self.device),
-            slot_mapping=torch.tensor(slots, dtype=torch.int64,
-                                      device=self.device),
-            is_decode=True,
-            block_tables=bt.to(self.device),
-            seq_lens=torch.tensor(seq_lens, dtype=torch.int32,
-                                  device=self.device))
+            input_ids=st.last_tok.index_select(0, rows_t),
+            positions=pos, slot_mapping=slot, is_decode=True,
+            block_tables=bt_rows, seq_lens=new_len)
         logits = self.model.forward(batch, self.pool.tensor)
-        tokens = self._sample(logits, [r.temperature for r in active])
+        any_temp = any(r.temperature > 0 for r in active)
+        tok = self._sample_device(logits, rows_t, any_temp)
+        st.last_tok.index_copy_(0, rows_t, tok)
+
+        n = tok.shape[0]
+        prev = self._collect_pending()     # read N-1 BEFORE reusing buffer
+        if self._cuda:
+            if self._pin is None or self._pin.shape[0] < st.capacity:
+                self._pin = torch.empty(st.capacity, dtype=torch.int64,
+                                        pin_memory=True)
+            self._pin[:n].copy_(tok, non_blocking=True)
+            ev = torch.cuda.Event()
+            ev.record()
+            self._pending = (active, ev, n)
+        else:
+            self._pending = (active, None, tok)
+        return prev + finals
+
+    def _collect_pending(self) -> List[RequestOutput]:
+        if self._pending is None:
+            return []
+        reqs, ev, payload = self._pending
+        self._pending = None
+        if ev is not None:
+            ev.synchronize()
+            vals = self._pin[:payload].tolist()
+        else:
+            vals = payload.tolist()
         outputs: List[RequestOutput] = []
         now = time.time()
-        for req, tok in zip(active, tokens):
+        for req, tok in zip(reqs, vals):
+            if self._by_id.get(req.request_id) is not req:
+                continue                      # aborted while in flight
+            req.inflight -= 1
             req.generated.append(int(tok))
             self.total_generated += 1
             if req.slo_ok:
@@ -356,13 +527,52 @@ class EngineWorker:
                     out.tpot_ms = (now - req.first_token_t) * 1e3 / (n_gen - 1)
                 out.e2e_ms = (now - req.arrival_t) * 1e3
                 self.running.remove(req)
+                self.dstate.leave(req.request_id)
                 self.mgr.free(req.request_id)
                 self._by_id.pop(req.request_id, None)
             outputs.append(out)
         return outputs
 
+    def _finalize(self, req: EngineRequest) -> RequestOutput:
+        """Emit the finished output for a request with no in-flight token
+        (its full generation came from prefill, e.g. max_tokens=1)."""
+        now = time.time()
+        n_gen = len(req.generated)
+        out = RequestOutput(
+            request_id=req.request_id, new_tokens=[], finished=True,
+            all_tokens=list(req.generated), prompt_tokens=req.prompt_len,
+            completion_tokens=n_gen, cached_tokens=req.cached_tokens,
+            ttft_ms=(req.first_token_t - req.arrival_t) * 1e3,
+            e2e_ms=(now - req.arrival_t) * 1e3)
+        if n_gen > 1:
+            out.tpot_ms = (now - req.first_token_t) * 1e3 / (n_gen - 1)
+        self.running.remove(req)
+        self.dstate.leave(req.request_id)
+        self.mgr.free(req.request_id)
+        self._by_id.pop(req.request_id, None)
+        return out
+
     # ---- sampling ----
-    def _sample(self, logits: torch.Tensor, temps: List[float]) -> List[int]:
+    def _sample_device(self, logits: torch.Tensor, rows_t: torch.Tensor,
+                       any_temp: bool) -> torch.Tensor:
+        """Greedy argmax, or gumbel-max for temperature>0 rows — entirely on
+        device (no host sync)."""
+        if not any_temp:
+            return logits.argmax(dim=-1)
+        temps = self.dstate.temps.index_select(0, rows_t)
+        hot = temps > 0
+        inv = torch.where(hot, 1.0 / temps.clamp(min=1e-6),
+                          torch.ones_like(temps))
+        y = logits.float() * inv.unsqueeze(1)
+        u = torch.rand_like(y).clamp_min(1e-20)
+        gumbel = -torch.log(-torch.log(u).clamp_min(1e-20))
+        y = y + gumbel * hot.unsqueeze(1)
+        return y.argmax(dim=-1)
+
+    def _sample_host(self, logits: torch.Tensor,
+                     temps: List[float]) -> List[int]:
+        """Prefill first-token sampling (host-visible; prefill already
+        syncs for TTFT bookkeeping)."""
         if logits.numel() == 0:
             return []
         greedy = logits.argmax(dim=-1)

@@ -4,9 +4,9 @@
 // whole point is streaming each sequence's K/V exactly once per step at
 // near-HBM rate (guide Appendix B "Attention decode").
 //
-// Geometry: one workgroup (4 waves, 256 threads) per (sequence, kv_head).
-// Q-heads in the GQA group (QPG = n_q_heads / n_kv_heads, <= 8) are scored
-// together so K rows are read once for the whole group.
+// Geometry: one workgroup (4 waves, 256 threads) per (sequence, kv_head,
+// partition). Q-heads in the GQA group (QPG = n_q_heads / n_kv_heads, <= 8)
+// are scored together so K rows are read once for the whole group.
 //   Phase A  each wave owns 64 tokens of a 256-token chunk; a lane streams
 //            its token's K row (short8 loads) against the group's Q vectors
 //            staged in LDS (broadcast reads) -> logits in LDS.
@@ -14,6 +14,12 @@
 //   Phase C  V accumulation: lane owns a dim pair, wave owns its 64 tokens;
 //            V rows stream fully coalesced (64 lanes x 4 B = one 256 B row).
 //   Final    cross-wave combine via LDS, normalize, bf16 store.
+//
+// Flash-decoding sequence split (SPLIT=true): long sequences are cut into
+// gridDim.z partitions so the launch has >> 256 workgroups (the chip needs
+// ~2 WGs/CU just to be full; B*KVH alone is often < 512). Each partition
+// writes an UNNORMALIZED partial (o_acc, m, l) and a tiny combine kernel
+// merges partitions: m* = max m_p; o = sum o_p*exp(m_p-m*); l likewise.
 #include "hip_common.h"
 
 namespace {
@@ -23,7 +29,7 @@ constexpr int CHUNK = 256;    // tokens per online-softmax chunk
 constexpr int NW = 4;         // waves per workgroup
 constexpr float NEG = -1e30f;
 
-template <int QPG>
+template <int QPG, bool SPLIT>
 __global__ __launch_bounds__(NW * WAVE) void paged_attention_kernel(
     const short* __restrict__ q,        // [B, QH, D]
     const short* __restrict__ k_cache,  // [NB, KVH, BS, D]
@@ -31,12 +37,18 @@ __global__ __launch_bounds__(NW * WAVE) void paged_attention_kernel(
     const int32_t* __restrict__ block_tables,  // [B, max_blocks]
     const int32_t* __restrict__ seq_lens,      // [B]
     short* __restrict__ out,                   // [B, QH, D]
-    int kvh, int bs, int max_blocks, float scale) {
+    float* __restrict__ part_o,   // [B, KVH, NP, QPG, D] (SPLIT)
+    float* __restrict__ part_ml,  // [B, KVH, NP, QPG, 2] (SPLIT)
+    int kvh, int bs, int max_blocks, int part_tokens, float scale) {
   const int b = blockIdx.x;
   const int kh = blockIdx.y;
+  const int part = SPLIT ? blockIdx.z : 0;
+  const int np = SPLIT ? gridDim.z : 1;
   const int qh0 = kh * QPG;
   const int n_q_heads = kvh * QPG;
   const int seq_len = seq_lens[b];
+  const int t_begin = SPLIT ? part * part_tokens : 0;
+  const int t_end = SPLIT ? min(seq_len, t_begin + part_tokens) : seq_len;
 
   __shared__ float q_lds[QPG][D];
   __shared__ float logits[QPG][CHUNK];
@@ -46,6 +58,17 @@ __global__ __launch_bounds__(NW * WAVE) void paged_attention_kernel(
   const int tid = threadIdx.x;
   const int wave = tid / WAVE;
   const int lane = tid % WAVE;
+
+  if (SPLIT && t_begin >= seq_len) {
+    // empty partition: publish l=0 so the combiner skips it
+    if (tid < QPG) {
+      float* ml = part_ml + ((((int64_t)b * kvh + kh) * np + part) * QPG +
+                             tid) * 2;
+      ml[0] = NEG;
+      ml[1] = 0.f;
+    }
+    return;
+  }
 
   // stage scaled Q for the group into LDS
   for (int i = tid; i < QPG * D; i += NW * WAVE) {
@@ -65,8 +88,8 @@ __global__ __launch_bounds__(NW * WAVE) void paged_attention_kernel(
 
   const int32_t* bt = block_tables + (int64_t)b * max_blocks;
 
-  for (int chunk0 = 0; chunk0 < seq_len; chunk0 += CHUNK) {
-    const int n_t = min(CHUNK, seq_len - chunk0);
+  for (int chunk0 = t_begin; chunk0 < t_end; chunk0 += CHUNK) {
+    const int n_t = min(CHUNK, t_end - chunk0);
     // ---- Phase A: logits[h][t_local] ----
     {
       const int t_local = wave * WAVE + lane;
@@ -165,15 +188,75 @@ __global__ __launch_bounds__(NW * WAVE) void paged_attention_kernel(
     comb[wave][h][2 * lane + 1] = o_acc[h][1];
   }
   __syncthreads();
-  // 256 threads cover QPG*D outputs (QPG<=8 -> <=1024 values, loop)
-  for (int i = tid; i < QPG * D; i += NW * WAVE) {
-    const int h = i / D, d = i % D;
-    float s = 0.f;
+  if (SPLIT) {
+    float* po = part_o + ((((int64_t)b * kvh + kh) * np + part) * QPG) * D;
+    for (int i = tid; i < QPG * D; i += NW * WAVE) {
+      const int h = i / D, d = i % D;
+      float s = 0.f;
 #pragma unroll
-    for (int w = 0; w < NW; ++w) s += comb[w][h][d];
-    const float l = l_sh[h];
-    out[((int64_t)b * n_q_heads + qh0 + h) * D + d] =
-        f32_to_bf16(l > 0.f ? s / l : 0.f);
+      for (int w = 0; w < NW; ++w) s += comb[w][h][d];
+      po[h * D + d] = s;                     // unnormalized
+    }
+    if (tid < QPG) {
+      float* ml = part_ml + ((((int64_t)b * kvh + kh) * np + part) * QPG +
+                             tid) * 2;
+      ml[0] = m_sh[tid];
+      ml[1] = l_sh[tid];
+    }
+  } else {
+    // 256 threads cover QPG*D outputs (QPG<=8 -> <=1024 values, loop)
+    for (int i = tid; i < QPG * D; i += NW * WAVE) {
+      const int h = i / D, d = i % D;
+      float s = 0.f;
+#pragma unroll
+      for (int w = 0; w < NW; ++w) s += comb[w][h][d];
+      const float l = l_sh[h];
+      out[((int64_t)b * n_q_heads + qh0 + h) * D + d] =
+          f32_to_bf16(l > 0.f ? s / l : 0.f);
+    }
+  }
+}
+
+// Merge the NP partitions of one (seq, kv_head): one workgroup per
+// (seq, kv_head), each thread owns (head, dim) outputs striding QPG*D.
+template <int QPG>
+__global__ __launch_bounds__(256) void paged_attention_combine_kernel(
+    const float* __restrict__ part_o,   // [B, KVH, NP, QPG, D]
+    const float* __restrict__ part_ml,  // [B, KVH, NP, QPG, 2]
+    short* __restrict__ out,            // [B, QH, D]
+    int kvh, int np) {
+  const int b = blockIdx.x;
+  const int kh = blockIdx.y;
+  const int n_q_heads = kvh * QPG;
+  __shared__ float m_g[QPG], scale_p[64][QPG];  // np <= 64
+
+  const int tid = threadIdx.x;
+  if (tid < QPG) {
+    float m = NEG;
+    for (int p = 0; p < np; ++p)
+      m = fmaxf(m, part_ml[((((int64_t)b * kvh + kh) * np + p) * QPG + tid)
+                           * 2]);
+    m_g[tid] = m;
+  }
+  __syncthreads();
+  for (int i = tid; i < QPG * np; i += 256) {
+    const int p = i / QPG, h = i % QPG;
+    const float* ml =
+        part_ml + ((((int64_t)b * kvh + kh) * np + p) * QPG + h) * 2;
+    scale_p[p][h] = (ml[1] > 0.f) ? __expf(ml[0] - m_g[h]) : 0.f;
+  }
+  __syncthreads();
+  for (int i = tid; i < QPG * D; i += 256) {
+    const int h = i / D, d = i % D;
+    float o = 0.f, l = 0.f;
+    for (int p = 0; p < np; ++p) {
+      const float s = scale_p[p][h];
+      if (s == 0.f) continue;
+      o += s * part_o[((((int64_t)b * kvh + kh) * np + p) * QPG + h) * D + d];
+      l += s * part_ml[((((int64_t)b * kvh + kh) * np + p) * QPG + h) * 2 + 1];
+    }
+    out[((int64_t)b * n_q_heads + kh * QPG + h) * D + d] =
+        f32_to_bf16(l > 0.f ? o / l : 0.f);
   }
 }
 
@@ -192,10 +275,45 @@ hipError_t lds_paged_attention(const void* q, const void* k_cache,
   const int qpg = n_q_heads / kvh;
   dim3 grid(n_seqs, kvh), block(NW * WAVE);
 #define LAUNCH(QPG)                                                           \
-  hipLaunchKernelGGL(paged_attention_kernel<QPG>, grid, block, 0, stream,     \
-                     (const short*)q, (const short*)k_cache,                  \
+  hipLaunchKernelGGL((paged_attention_kernel<QPG, false>), grid, block, 0,    \
+                     stream, (const short*)q, (const short*)k_cache,          \
                      (const short*)v_cache, block_tables, seq_lens,           \
-                     (short*)out, kvh, bs, max_blocks, scale)
+                     (short*)out, nullptr, nullptr, kvh, bs, max_blocks, 0,   \
+                     scale)
+  switch (qpg) {
+    case 1: LAUNCH(1); break;
+    case 2: LAUNCH(2); break;
+    case 4: LAUNCH(4); break;
+    case 8: LAUNCH(8); break;
+    default: return hipErrorInvalidValue;
+  }
+#undef LAUNCH
+  HIP_CHECK_LAST();
+  return hipSuccess;
+}
+
+hipError_t lds_paged_attention_split(
+    const void* q, const void* k_cache, const void* v_cache,
+    const int32_t* block_tables, const int32_t* seq_lens, void* out,
+    float* part_o, float* part_ml, int n_seqs, int n_q_heads, int kvh, int bs,
+    int head_dim, int max_blocks, int n_parts, int part_tokens, float scale,
+    hipStream_t stream) {
+  if (n_seqs == 0) return hipSuccess;
+  if (head_dim != D || n_parts > 64) return hipErrorInvalidValue;
+  const int qpg = n_q_heads / kvh;
+  dim3 grid(n_seqs, kvh, n_parts), block(NW * WAVE);
+  dim3 cgrid(n_seqs, kvh), cblock(256);
+#define LAUNCH(QPG)                                                           \
+  do {                                                                        \
+    hipLaunchKernelGGL((paged_attention_kernel<QPG, true>), grid, block, 0,   \
+                       stream, (const short*)q, (const short*)k_cache,        \
+                       (const short*)v_cache, block_tables, seq_lens,         \
+                       (short*)out, part_o, part_ml, kvh, bs, max_blocks,     \
+                       part_tokens, scale);                                   \
+    hipLaunchKernelGGL((paged_attention_combine_kernel<QPG>), cgrid, cblock,  \
+                       0, stream, part_o, part_ml, (short*)out, kvh,          \
+                       n_parts);                                              \
+  } while (0)
   switch (qpg) {
     case 1: LAUNCH(1); break;
     case 2: LAUNCH(2); break;

@@ -18,6 +18,7 @@ import time
 from dataclasses import dataclass, field
 from typing import Dict, List, Optional
 
+import numpy as np
 import torch
 
 from ..datalayer.endpoint import Metrics, Role
@@ -303,8 +304,10 @@ class EngineWorker:
         embed_rows, embed_vals = [], []
         finishing: List[EngineRequest] = []
         bs = self.pool.block_size
+        metas = []                      # (seq_start, chunk, prior) per seq
         for req, chunk in selected:
             start, end = req.computed, req.computed + chunk
+            metas.append((seq_starts[-1], chunk, start))
             if req.prefix_embeds is not None and start < req.prefix_embeds.shape[0]:
                 e_end = min(end, req.prefix_embeds.shape[0])
                 base_row = seq_starts[-1] + 0
@@ -315,8 +318,7 @@ class EngineWorker:
             slots.extend(self.mgr.slots_for_range(req.request_id, start, end))
             self.mgr.set_seq_len(req.request_id, end)
             ctx_lens.append(end)
-            tables.append(torch.tensor(self.mgr.tables[req.request_id],
-                                       dtype=torch.int32, device=self.device))
+            tables.append(self.mgr.tables[req.request_id])
             seq_starts.append(seq_starts[-1] + chunk)
             if end == req.prompt_len:
                 logit_rows.append(seq_starts[-1] - 1)
@@ -331,6 +333,28 @@ class EngineWorker:
                                         int(req.block_hashes[b]))
             req.registered_blocks = max(req.registered_blocks, full)
 
+        qpg = self.cfg.num_heads // self.cfg.num_kv_heads
+        flash_ok = (self._cuda and self.cfg.head_dim == 128
+                    and qpg in (1, 2, 4, 8)
+                    and self.dtype == torch.bfloat16)
+        prefill_bt = prefill_meta = prefill_tiles = None
+        bt_list = None
+        if flash_ok:
+            maxb = max(len(t) for t in tables)
+            bt_np = np.zeros((len(tables), maxb), dtype=np.int32)
+            for i, t in enumerate(tables):
+                bt_np[i, :len(t)] = t
+            prefill_bt = torch.from_numpy(bt_np).to(self.device,
+                                                    non_blocking=True)
+            prefill_meta = torch.tensor(metas, dtype=torch.int32).to(
+                self.device, non_blocking=True)
+            tl = [(i, v0) for i, (_, chunk, _) in enumerate(metas)
+                  for v0 in range(0, chunk * qpg, 128)]
+            prefill_tiles = torch.tensor(tl, dtype=torch.int32).to(
+                self.device, non_blocking=True)
+        else:
+            bt_list = [torch.tensor(t, dtype=torch.int32,
+                                    device=self.device) for t in tables]
         batch = ForwardBatch(
             input_ids=torch.tensor(input_ids, dtype=torch.int64,
                                    device=self.device),
@@ -339,7 +363,9 @@ class EngineWorker:
             slot_mapping=torch.tensor(slots, dtype=torch.int64,
                                       device=self.device),
             is_decode=False, seq_starts=seq_starts, ctx_lens=ctx_lens,
-            prefill_block_tables=tables,
+            prefill_block_tables=bt_list,
+            prefill_bt=prefill_bt, prefill_meta=prefill_meta,
+            prefill_tiles=prefill_tiles,
             logit_rows=torch.tensor(logit_rows, dtype=torch.int64,
                                     device=self.device)
             if logit_rows else torch.zeros(0, dtype=torch.int64,
@@ -473,7 +499,9 @@ class EngineWorker:
         batch = ForwardBatch(
             input_ids=st.last_tok.index_select(0, rows_t),
             positions=pos, slot_mapping=slot, is_decode=True,
-            block_tables=bt_rows, seq_lens=new_len)
+            block_tables=bt_rows, seq_lens=new_len,
+            max_seq_len=max(self.mgr.seq_lens[r.request_id]
+                            for r in active))
         logits = self.model.forward(batch, self.pool.tensor)
         any_temp = any(r.temperature > 0 for r in active)
         tok = self._sample_device(logits, rows_t, any_temp)

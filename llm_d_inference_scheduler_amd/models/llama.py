@@ -36,10 +36,15 @@ class ForwardBatch:
     # decode-only:
     block_tables: Optional[torch.Tensor] = None   # [B, max_blocks] int32
     seq_lens: Optional[torch.Tensor] = None       # [B] int32 (incl. new token)
+    max_seq_len: Optional[int] = None             # host-known max (no sync)
     # prefill-only:
     seq_starts: Optional[List[int]] = None        # len B+1 offsets into T
     ctx_lens: Optional[List[int]] = None          # [B] total ctx after chunk
     prefill_block_tables: Optional[List[torch.Tensor]] = None
+    # fused flash-prefill metadata (GPU path; see flash_prefill.hip)
+    prefill_bt: Optional[torch.Tensor] = None     # [S, max_blocks] int32
+    prefill_meta: Optional[torch.Tensor] = None   # [S, 3] start,chunk,prior
+    prefill_tiles: Optional[torch.Tensor] = None  # [n_tiles, 2] seq,vrow0
     logit_rows: Optional[torch.Tensor] = None     # rows needing logits
     # multimodal: rows whose input embedding is provided (encode hand-off)
     embed_rows: Optional[torch.Tensor] = None     # [n] int64 into T
@@ -131,7 +136,8 @@ class LlamaRunner:
             if batch.is_decode:
                 attn = ops.paged_attention(q, k_cache, v_cache,
                                            batch.block_tables, batch.seq_lens,
-                                           self.scale)
+                                           self.scale,
+                                           max_seq_len=batch.max_seq_len)
             else:
                 attn = self._prefill_attention(batch, q, k_cache, v_cache)
             o = attn.view(T, c.q_size) @ layer["wo"]
@@ -152,9 +158,16 @@ class LlamaRunner:
     def _prefill_attention(self, batch: ForwardBatch, q: torch.Tensor,
                            k_cache: torch.Tensor,
                            v_cache: torch.Tensor) -> torch.Tensor:
-        """Causal varlen attention per sequence over pool-gathered KV.
-        Chunked over query rows to bound the score matrix."""
+        """Causal varlen attention per sequence. GPU: fused MFMA flash
+        kernel (flash_prefill.hip) — S never materialized. CPU fallback:
+        composed chunked torch attention over pool-gathered KV."""
         c = self.cfg
+        if (q.is_cuda and batch.prefill_tiles is not None
+                and c.head_dim == 128
+                and c.num_heads // c.num_kv_heads in (1, 2, 4, 8)):
+            return ops.flash_prefill(
+                q, k_cache, v_cache, batch.prefill_bt, batch.prefill_meta,
+                batch.prefill_tiles, self.scale)
         qpg = c.num_heads // c.num_kv_heads
         out = torch.empty_like(q)
         starts = batch.seq_starts

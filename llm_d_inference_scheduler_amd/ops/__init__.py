@@ -1,4 +1,4 @@
 from .dispatch import (  # noqa: F401
-    hip_ops, move_blocks, paged_attention, reshape_and_cache, rmsnorm, rope,
-    silu_mul,
+    flash_prefill, hip_ops, move_blocks, paged_attention, reshape_and_cache,
+    rmsnorm, rope, silu_mul,
 )

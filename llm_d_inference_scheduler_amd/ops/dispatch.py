@@ -67,13 +67,39 @@ def reshape_and_cache(k_new: torch.Tensor, v_new: torch.Tensor,
 
 def paged_attention(q: torch.Tensor, k_cache: torch.Tensor,
                     v_cache: torch.Tensor, block_tables: torch.Tensor,
-                    seq_lens: torch.Tensor, scale: float) -> torch.Tensor:
+                    seq_lens: torch.Tensor, scale: float,
+                    max_seq_len: Optional[int] = None) -> torch.Tensor:
     if _use_hip(q):
-        return _ext.paged_attention(q, k_cache, v_cache,
-                                    block_tables.to(torch.int32),
-                                    seq_lens.to(torch.int32), scale)
+        bt = block_tables.to(torch.int32)
+        sl = seq_lens.to(torch.int32)
+        # flash-decoding sequence split: the plain kernel launches only
+        # B*KVH workgroups — far short of the 256-CU chip's >=2 WG/CU need.
+        # Partition long sequences so the grid fills the chip; a combine
+        # kernel merges the unnormalized partials.
+        n_wgs = q.shape[0] * k_cache.shape[1]
+        if max_seq_len is not None and n_wgs < 1024 and max_seq_len > 512:
+            target_np = min(64, max(1, 1024 // max(1, n_wgs)))
+            # partition size: ceil-divide then round up to the kernel's
+            # 256-token online-softmax chunk
+            part = max(256, ((max_seq_len + target_np - 1) // target_np
+                             + 255) // 256 * 256)
+            np_ = (max_seq_len + part - 1) // part
+            if np_ > 1:
+                return _ext.paged_attention_split(q, k_cache, v_cache, bt,
+                                                  sl, np_, part, scale)
+        return _ext.paged_attention(q, k_cache, v_cache, bt, sl, scale)
     return ref.paged_attention(q, k_cache, v_cache, block_tables, seq_lens,
                                scale)
+
+
+def flash_prefill(q: torch.Tensor, k_cache: torch.Tensor,
+                  v_cache: torch.Tensor, block_tables: torch.Tensor,
+                  seq_meta: torch.Tensor, tiles: torch.Tensor,
+                  scale: float) -> torch.Tensor:
+    """Fused MFMA causal varlen prefill attention (GPU-only; callers fall
+    back to the composed reference path when unsupported)."""
+    return hip_ops().flash_prefill(q, k_cache, v_cache, block_tables,
+                                   seq_meta, tiles, scale)
 
 
 def move_blocks(pool: torch.Tensor, staging: torch.Tensor,

@@ -172,6 +172,92 @@ class TestEngineKernelsGPU:
         assert torch.allclose(to_f32(out), out_ref, atol=3e-2, rtol=3e-2), \
             (to_f32(out) - out_ref).abs().max()
 
+    @pytest.mark.parametrize("seqs,np_,part", [([1024, 2049, 333], 4, 768),
+                                               ([8192], 16, 512),
+                                               ([100, 700], 2, 512)])
+    def test_paged_attention_split(self, ext, seqs, np_, part):
+        """Flash-decoding partitioned kernel == the single-pass kernel."""
+        torch.manual_seed(6)
+        KVH, D, BS, qpg = 8, 128, 16, 4
+        QH = KVH * qpg
+        B = len(seqs)
+        max_blocks = (max(seqs) + BS - 1) // BS
+        NB = max_blocks * B + 1
+        q = torch.randn(B, QH, D, device="cuda").bfloat16()
+        kc = torch.randn(NB, KVH, BS, D, device="cuda").bfloat16()
+        vc = torch.randn(NB, KVH, BS, D, device="cuda").bfloat16()
+        bt = torch.zeros(B, max_blocks, dtype=torch.int32)
+        perm = torch.randperm(NB - 1) + 1
+        k = 0
+        for b, s in enumerate(seqs):
+            nb = (s + BS - 1) // BS
+            bt[b, :nb] = perm[k:k + nb]
+            k += nb
+        sl = torch.tensor(seqs, dtype=torch.int32)
+        base = ext.paged_attention(q, kc, vc, bt.cuda(), sl.cuda(), D ** -0.5)
+        split = ext.paged_attention_split(q, kc, vc, bt.cuda(), sl.cuda(),
+                                          np_, part, D ** -0.5)
+        assert torch.allclose(to_f32(base), to_f32(split), atol=2e-2,
+                              rtol=2e-2), (to_f32(base) - to_f32(split)).abs().max()
+
+    @pytest.mark.parametrize(
+        "qpg,chunks,priors",
+        [(4, [128, 333, 1024], [0, 171, 0]),     # mixed varlen + chunked
+         (8, [257], [512]),                      # chunked continuation
+         (1, [64], [0]),                         # MHA (LLaVA text shape)
+         (4, [2048], [0])])                      # long single prefill
+    def test_flash_prefill(self, ext, qpg, chunks, priors):
+        """Fused MFMA flash prefill vs fp32 composed causal attention."""
+        torch.manual_seed(7)
+        KVH, D, BS = 8, 128, 16
+        QH = KVH * qpg
+        ctxs = [c + p for c, p in zip(chunks, priors)]
+        max_blocks = (max(ctxs) + BS - 1) // BS
+        NB = max_blocks * len(chunks) + 1
+        T = sum(chunks)
+        q = torch.randn(T, QH, D, device="cuda").bfloat16()
+        kc = torch.randn(NB, KVH, BS, D, device="cuda").bfloat16()
+        vc = torch.randn(NB, KVH, BS, D, device="cuda").bfloat16()
+        bt = torch.zeros(len(chunks), max_blocks, dtype=torch.int32)
+        perm = torch.randperm(NB - 1) + 1
+        k = 0
+        starts = [0]
+        for i, ctx in enumerate(ctxs):
+            nb = (ctx + BS - 1) // BS
+            bt[i, :nb] = perm[k:k + nb]
+            k += nb
+            starts.append(starts[-1] + chunks[i])
+        metas = [(starts[i], chunks[i], priors[i])
+                 for i in range(len(chunks))]
+        tiles = [(i, v0) for i in range(len(chunks))
+                 for v0 in range(0, chunks[i] * qpg, 128)]
+        scale = D ** -0.5
+        out = ext.flash_prefill(
+            q, kc, vc, bt.cuda(),
+            torch.tensor(metas, dtype=torch.int32, device="cuda"),
+            torch.tensor(tiles, dtype=torch.int32, device="cuda"), scale)
+        # fp32 reference: gather each sequence's KV, causal softmax
+        out_ref = torch.empty(T, QH, D)
+        qf = to_f32(q)
+        kcf, vcf = to_f32(kc), to_f32(vc)
+        for i, ctx in enumerate(ctxs):
+            rows = bt[i, :(ctx + BS - 1) // BS].long()
+            kk = kcf[rows].permute(1, 0, 2, 3).reshape(KVH, -1, D)[:, :ctx]
+            vv = vcf[rows].permute(1, 0, 2, 3).reshape(KVH, -1, D)[:, :ctx]
+            qi = qf[starts[i]:starts[i + 1]].view(
+                chunks[i], KVH, qpg, D).permute(1, 2, 0, 3)
+            s = torch.einsum("hgtd,hsd->hgts", qi, kk) * scale
+            t_idx = torch.arange(chunks[i]).view(1, 1, -1, 1)
+            s_idx = torch.arange(ctx).view(1, 1, 1, -1)
+            s.masked_fill_(s_idx > t_idx + priors[i], float("-inf"))
+            p = torch.softmax(s, dim=-1)
+            o = torch.einsum("hgts,hsd->hgtd", p, vv)
+            out_ref[starts[i]:starts[i + 1]] = o.permute(2, 0, 1, 3).reshape(
+                chunks[i], QH, D)
+        diff = (to_f32(out) - out_ref).abs().max()
+        assert torch.allclose(to_f32(out), out_ref, atol=3e-2,
+                              rtol=3e-2), diff
+
     def test_move_blocks_roundtrip(self, ext):
         L, NB, KVH, BS, D = 4, 64, 8, 16, 128
         pool = torch.randn(L, 2, NB, KVH, BS, D, device="cuda").bfloat16()

@@ -155,18 +155,25 @@ __global__ __launch_bounds__(NW * WAVE) void flash_prefill_kernel(
     }
     tmax = fmaxf(tmax, __shfl_xor(tmax, 32, WAVE));
     const float m_new = fmaxf(m_run, tmax);
-    if (m_new > NEG * 0.5f) {
-      const float alpha = (m_run <= NEG * 0.5f) ? 0.f : __expf(m_run - m_new);
+    // MFMA reads operand registers from ALL 64 lanes regardless of EXEC,
+    // so the PV block below must run wave-uniformly: lanes with no valid
+    // element yet (m_new still NEG) contribute zero P columns instead of
+    // branching around the matrix op (a per-lane `if` here corrupts the
+    // shared A/B operands with uninitialized registers — measured).
+    const bool has = m_new > NEG * 0.5f;
+    if (__any(has)) {
+      const float m_eff = has ? m_new : 0.f;
+      const float alpha = (m_run <= NEG * 0.5f) ? 0.f : __expf(m_run - m_eff);
       float tsum = 0.f;
       float pr[16];
 #pragma unroll
       for (int r = 0; r < 16; ++r) {
-        pr[r] = (s[r] <= NEG * 0.5f) ? 0.f : __expf(s[r] - m_new);
+        pr[r] = (s[r] <= NEG * 0.5f) ? 0.f : __expf(s[r] - m_eff);
         tsum += pr[r];
       }
       tsum += __shfl_xor(tsum, 32, WAVE);
       l_run = l_run * alpha + tsum;
-      m_run = m_new;
+      if (has) m_run = m_new;
 #pragma unroll
       for (int dt = 0; dt < D / 32; ++dt)
 #pragma unroll

@@ -236,11 +236,17 @@ def main():
     stats = {"completed": 0, "errors": 0, "epp_ms": [], "ttft_ms": [],
              "e2e_ms": []}
 
-    def feed():
+    ramp = max(1, target_inflight // max(8, args.warmup // 2))
+
+    def feed(limit=None):
         if rank != 0:
             return
+        n = 0
         while node.inflight + len(node._arrivals) < target_inflight:
+            if limit is not None and n >= limit:
+                break
             node.submit(workload.next_request())
+            n += 1
 
     def drain():
         if rank != 0:
@@ -263,9 +269,11 @@ def main():
             if use_gpu:
                 torch.cuda.synchronize()
 
-    # ---- warmup ----
+    # ---- warmup (ramped admission: the closed-loop cohort arrives over
+    # the first half of warmup instead of as one thundering herd whose
+    # tail TTFTs blow the SLO before steady state exists) ----
     for _ in range(args.warmup):
-        feed()
+        feed(limit=ramp)
         node.step()
         drain()
     if rank == 0:

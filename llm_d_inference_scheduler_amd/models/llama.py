@@ -12,6 +12,7 @@ The reference router never runs a model — its pods do (vLLM). This runner
 is the per-GPU worker engine's model, SURVEY.md §2.12 "worker shim".
 """
 import math
+import os
 from dataclasses import dataclass
 from typing import Dict, List, Optional
 
@@ -19,7 +20,30 @@ import torch
 
 from .. import ops
 from ..ops import ref as ops_ref
+from ..utils.logging import get_logger
 from .configs import ModelConfig
+
+log = get_logger("models.llama")
+
+_TUNED_GEMM_LOADED = False
+
+
+def _load_tuned_gemms() -> None:
+    """Load offline hipBLASLt autotune results (tools/tune_gemms.py) so the
+    skinny decode GEMMs use tuned solutions instead of the default pick
+    (~2.6x off the weights-bound floor, profiles/r01_bench_kernel_stats_v3)."""
+    global _TUNED_GEMM_LOADED
+    if _TUNED_GEMM_LOADED:
+        return
+    _TUNED_GEMM_LOADED = True
+    path = os.path.join(os.path.dirname(os.path.abspath(__file__)),
+                        "..", "..", "profiles", "tuned_gemm_gfx950.csv")
+    if os.path.exists(path):
+        import torch.cuda.tunable as tunable
+        tunable.enable(True)
+        tunable.tuning_enable(False)
+        tunable.read_file(path)
+        log.info("loaded tuned GEMM solutions from %s", path)
 
 
 @dataclass
@@ -60,6 +84,7 @@ class LlamaRunner:
         self.scale = 1.0 / math.sqrt(config.head_dim)
         c = config
         if self.device.type == "cuda":
+            _load_tuned_gemms()
             # init directly on the GPU (16 GB of weights for Llama-3-8B —
             # CPU-side init would dominate startup). Same seed => identical
             # weights on every rank (required by the P/D KV hand-off).

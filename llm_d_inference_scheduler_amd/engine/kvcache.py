@@ -88,6 +88,10 @@ class BlockManager:
         # stats
         self.cached_tokens_total = 0
         self.queried_tokens_total = 0
+        # KV event stream for the router's precise prefix index
+        # (replaces vLLM's ZMQ KV events, precise_prefix_cache.go:655-671)
+        self._ev_stored: List[int] = []
+        self._ev_evicted: List[int] = []
 
     @property
     def free_blocks(self) -> int:
@@ -109,6 +113,7 @@ class BlockManager:
         if h is not None:                      # evict stale cached content
             if self._by_hash.get(h) == blk:
                 del self._by_hash[h]
+                self._ev_evicted.append(h)
             self._hash_of[blk] = None
         self._refcnt[blk] = 1
         return blk
@@ -159,6 +164,7 @@ class BlockManager:
             return                              # first writer wins
         self._by_hash[h] = blk
         self._hash_of[blk] = h
+        self._ev_stored.append(h)
 
     # ---- allocation ------------------------------------------------------
     def can_allocate(self, n_tokens: int) -> bool:
@@ -212,3 +218,9 @@ class BlockManager:
     def release_blocks(self, blocks: List[int]) -> None:
         for blk in blocks:
             self._release(blk)
+
+    def drain_events(self):
+        """(stored, evicted) content-hash batches since the last drain."""
+        s, e = self._ev_stored, self._ev_evicted
+        self._ev_stored, self._ev_evicted = [], []
+        return s, e

@@ -185,9 +185,13 @@ class NodeRunner:
         # in two kernel launches on the router rank's GPU
         self._approx = None
         self._gpu_prefix = None
+        self._precise = None
+        from ..plugins.scorers import PrecisePrefixCacheScorer
         for p in self.loaded.plugins.values():
             if isinstance(p, ApproxPrefixCacheProducer):
                 self._approx = p
+            elif isinstance(p, PrecisePrefixCacheScorer):
+                self._precise = p
         if self._approx is not None and str(cfg.device).startswith("cuda"):
             from ..ops.prefix import GpuPrefixIndex
             self._gpu_prefix = GpuPrefixIndex(cfg.device)
@@ -235,6 +239,12 @@ class NodeRunner:
             self._route_arrivals()
         self._outbox.append({"type": "metrics", "src": self.rank,
                              "m": self._metrics_payload()})
+        stored, evicted = self.engine.mgr.drain_events()
+        if stored or evicted:
+            # engine KV events for the precise prefix index (replaces the
+            # reference's per-pod ZMQ KV-event subscriptions)
+            self._outbox.append({"type": "kv_events", "src": self.rank,
+                                 "s": stored, "e": evicted})
         msgs = self.mailbox.exchange(self._outbox)
         self._outbox = []
         self._process_messages(msgs)
@@ -384,6 +394,10 @@ class NodeRunner:
                 self._handle_assign(m)
             elif t == "done" and self.is_router:
                 self._handle_done(m)
+            elif t == "kv_events" and self.is_router:
+                if self._precise is not None:
+                    self._precise.apply_events(f"gpu{m['src']}", m["s"],
+                                               m["e"])
             elif t == "tokens" and self.is_router:
                 self._token_events.append((m["req_id"], m["toks"]))
                 decision = self._decisions.get(m["req_id"])

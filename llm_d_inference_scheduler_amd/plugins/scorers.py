@@ -263,3 +263,56 @@ class LatencyScorer(Scorer):
             # squash headroom (ms) into [0,1]; 0 headroom -> 0.5
             out[ep.name] = 1.0 / (1.0 + math.exp(-h / 100.0))
         return out
+
+
+@register_plugin("precise-prefix-cache-scorer")
+class PrecisePrefixCacheScorer(Scorer):
+    """Exact KV-block index scorer (scorer/preciseprefixcache/
+    precise_prefix_cache.go). The reference subscribes to per-pod vLLM KV
+    events over ZMQ; here the engines' BlockManagers emit real content
+    hashes over the node mailbox into a KVBlockIndex (datalayer/kvblock.py),
+    plus 2s-TTL speculative entries written at PreRequest (:533-604).
+
+    Hashing uses the ENGINE's chained-block scheme (engine/kvcache.py
+    block_hashes) so scorer lookups match event hashes bit-for-bit.
+    """
+
+    def __init__(self, name: str = "", **params):
+        super().__init__(name, **params)
+        from ..datalayer.kvblock import KVBlockIndex
+        self.block_size = int(params.get("blockSize", 16))
+        self.index = KVBlockIndex(
+            float(params.get("speculativeTTLSeconds", 2.0)))
+
+    def _hashes(self, ctx: SchedulingContext):
+        cached = getattr(ctx, "_precise_hashes", None)
+        if cached is not None:
+            return cached
+        from ..engine.kvcache import block_hashes
+        toks = ctx.request.prompt_tokens or []
+        h = [int(x) for x in block_hashes(toks, self.block_size)]
+        ctx._precise_hashes = h
+        return h
+
+    def score(self, ctx: SchedulingContext, endpoints):
+        hashes = self._hashes(ctx)
+        if not hashes:
+            return {ep.name: 0.0 for ep in endpoints}
+        matched = self.index.match_longest(hashes,
+                                           [ep.name for ep in endpoints])
+        total = len(hashes)
+        return {name: n / total for name, n in matched.items()}
+
+    def pre_request(self, ctx: SchedulingContext, result, target) -> None:
+        if target is None:
+            return
+        hashes = self._hashes(ctx)
+        if hashes:
+            self.index.add_speculative(target.name, hashes)
+
+    def apply_events(self, endpoint_name: str, stored, evicted) -> None:
+        self.index.apply_events(endpoint_name, stored, evicted)
+        self.index.sweep()
+
+    def remove_endpoint(self, ep: Endpoint) -> None:
+        self.index.remove_endpoint(ep.name)

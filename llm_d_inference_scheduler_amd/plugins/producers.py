@@ -202,24 +202,89 @@ class _OnlineStat:
 class PredictedLatencyProducer(DataProducer):
     """TTFT/TPOT prediction + SLO headroom (dataproducer/predictedlatency).
 
-    The reference trains XGBoost/BayesianRidge in an external python sidecar
-    over HTTP; here the predictor is in-process: per-endpoint online stats
-      TTFT ~= base + queue_depth * step_ms + non_cached_tokens * per_token_ms
-      TPOT ~= base_tpot * (1 + running / batch_scale)
-    updated from response_complete samples.
+    The reference trains XGBoost/LightGBM/BayesianRidge in an external
+    python predictor sidecar over HTTP (predictedlatency/training.go,
+    latencypredictorclient/). Here the predictor is in-process, same
+    two-stage design:
+      * cold start: per-endpoint online stats
+          TTFT ~= queue_depth * step_ms + non_cached_tokens * per_token_ms
+          TPOT ~= base_tpot * (1 + running / batch_scale)
+      * trained: a BayesianRidge model per metric over routing-time
+        features [queue, running, kv_usage, non_cached_tokens, prompt_len],
+        retrained in-process every `retrainEvery` completed samples
+        (features captured at PreRequest, labels at response-complete —
+        the reference's training-sample collection hooks).
     """
 
     requires = ["token-producer", "approx-prefix-cache-producer"]
     produces = LATENCY_PREDICTION_INFO
 
+    N_FEATURES = 5
+
     def __init__(self, name: str = "", **params):
         super().__init__(name, **params)
         self.default_ttft_slo_ms = float(params.get("ttftSLOms", 2000.0))
         self.default_tpot_slo_ms = float(params.get("tpotSLOms", 100.0))
+        self.retrain_every = int(params.get("retrainEvery", 256))
+        self.max_samples = int(params.get("maxSamples", 4096))
         self._per_token = {}
         self._step_ms = {}
         self._tpot = {}
         self._lock = threading.Lock()
+        # trained-model state
+        from collections import deque
+        self._pending_feats: Dict[str, tuple] = {}   # req_id -> (ep, feats)
+        self._ttft_samples = deque(maxlen=self.max_samples)
+        self._tpot_samples = deque(maxlen=self.max_samples)
+        self._since_train = 0
+        self._ttft_model = None
+        self._tpot_model = None
+
+    @staticmethod
+    def _features(ep: Endpoint, non_cached: int, n_tokens: int):
+        m = ep.metrics
+        return [float(m.waiting_queue_size),
+                float(m.running_requests_size),
+                float(m.kv_cache_usage),
+                float(non_cached), float(n_tokens)]
+
+    def _maybe_train(self) -> None:
+        if self._since_train < self.retrain_every or \
+                len(self._ttft_samples) < 64:
+            return
+        self._since_train = 0
+        try:
+            from sklearn.linear_model import BayesianRidge
+        except ImportError:           # predictor degrades to online stats
+            return
+        import numpy as _np
+        xs, ys = zip(*self._ttft_samples)
+        m = BayesianRidge()
+        m.fit(_np.asarray(xs), _np.asarray(ys))
+        self._ttft_model = m
+        if len(self._tpot_samples) >= 64:
+            xs, ys = zip(*self._tpot_samples)
+            m2 = BayesianRidge()
+            m2.fit(_np.asarray(xs), _np.asarray(ys))
+            self._tpot_model = m2
+
+    def pre_request(self, ctx: SchedulingContext, result, target) -> None:
+        """Capture routing-time features for training-label pairing."""
+        if target is None:
+            return
+        req = ctx.request
+        n_tokens = len(req.prompt_tokens or []) or \
+            max(1, req.prompt_len_chars // 4)
+        prefix = ctx.attributes.get(PREFIX_CACHE_MATCH_INFO)
+        cached = 0
+        if prefix is not None:
+            cached = prefix.match_blocks.get(target.name, 0) * \
+                prefix.block_size_tokens
+        feats = self._features(target, max(0, n_tokens - cached), n_tokens)
+        with self._lock:
+            self._pending_feats[req.request_id] = (target.name, feats)
+            if len(self._pending_feats) > 8 * self.max_samples:
+                self._pending_feats.clear()    # leak guard on lost requests
 
     def _stats(self, name: str):
         with self._lock:
@@ -238,21 +303,40 @@ class PredictedLatencyProducer(DataProducer):
         info = LatencyPredictionInfo(
             ttft_slo_ms=req.ttft_slo_ms or self.default_ttft_slo_ms,
             tpot_slo_ms=req.tpot_slo_ms or self.default_tpot_slo_ms)
+        ttft_model, tpot_model = self._ttft_model, self._tpot_model
+        feats_all = []
         for ep in endpoints:
-            per_token, step_ms, tpot = self._stats(ep.name)
             cached = 0
             if prefix is not None:
                 cached = prefix.match_blocks.get(ep.name, 0) * \
                     prefix.block_size_tokens
-            non_cached = max(0, n_tokens - cached)
+            feats_all.append(self._features(ep, max(0, n_tokens - cached),
+                                            n_tokens))
+        if ttft_model is not None:
+            # bulk prediction across candidates (one predictor call per
+            # request, as the reference's batched HTTP client does)
+            import numpy as _np
+            x = _np.asarray(feats_all)
+            pred_ttfts = ttft_model.predict(x)
+            pred_tpots = (tpot_model.predict(x) if tpot_model is not None
+                          else [None] * len(endpoints))
+        else:
+            pred_ttfts = pred_tpots = [None] * len(endpoints)
+        for ep, feats, p_ttft, p_tpot in zip(endpoints, feats_all,
+                                             pred_ttfts, pred_tpots):
+            per_token, step_ms, tpot = self._stats(ep.name)
             m = ep.metrics
-            pred_ttft = (m.waiting_queue_size * step_ms.value
-                         + non_cached * per_token.value)
-            pred_tpot = tpot.value * (1.0 + m.running_requests_size / 64.0)
-            info.predicted_ttft_ms[ep.name] = pred_ttft
-            info.predicted_tpot_ms[ep.name] = pred_tpot
-            info.ttft_headroom_ms[ep.name] = info.ttft_slo_ms - pred_ttft
-            info.tpot_headroom_ms[ep.name] = info.tpot_slo_ms - pred_tpot
+            if p_ttft is None:
+                p_ttft = (m.waiting_queue_size * step_ms.value
+                          + feats[3] * per_token.value)
+            if p_tpot is None:
+                p_tpot = tpot.value * (1.0 + m.running_requests_size / 64.0)
+            p_ttft = max(0.0, float(p_ttft))
+            p_tpot = max(0.0, float(p_tpot))
+            info.predicted_ttft_ms[ep.name] = p_ttft
+            info.predicted_tpot_ms[ep.name] = p_tpot
+            info.ttft_headroom_ms[ep.name] = info.ttft_slo_ms - p_ttft
+            info.tpot_headroom_ms[ep.name] = info.tpot_slo_ms - p_tpot
         ctx.attributes[LATENCY_PREDICTION_INFO] = info
 
     def response_complete(self, ctx: SchedulingContext, target, usage) -> None:
@@ -268,3 +352,14 @@ class PredictedLatencyProducer(DataProducer):
         tpot_ms = getattr(usage, "tpot_ms", None)
         if tpot_ms is not None and tpot_ms > 0:
             tpot.update(tpot_ms)
+        # trained-model sample: pair routing-time features with the label
+        with self._lock:
+            rec = self._pending_feats.pop(ctx.request.request_id, None)
+            if rec is not None:
+                _, feats = rec
+                if ttft_ms is not None and ttft_ms > 0:
+                    self._ttft_samples.append((feats, float(ttft_ms)))
+                    self._since_train += 1
+                if tpot_ms is not None and tpot_ms > 0:
+                    self._tpot_samples.append((feats, float(tpot_ms)))
+            self._maybe_train()

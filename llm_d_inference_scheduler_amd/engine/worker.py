@@ -210,8 +210,17 @@ class EngineWorker:
 
     def admit_transferred(self, req: EngineRequest, local_blocks: List[int],
                           seq_len: int, first_token: int) -> None:
-        """Adopt a prefilled sequence whose KV arrived over xGMI."""
+        """Adopt a prefilled sequence whose KV arrived over xGMI. The
+        adopted full blocks are registered in the prefix cache: a
+        transferred prefix is as reusable as a locally-computed one."""
         self.mgr.adopt(req.request_id, local_blocks, seq_len)
+        if req.block_hashes is None:
+            req.block_hashes = block_hashes(req.prompt_tokens,
+                                            self.pool.block_size)
+        full = min(len(local_blocks), req.prompt_len // self.pool.block_size)
+        for b in range(full):
+            self.mgr.register_block(req.request_id, b,
+                                    int(req.block_hashes[b]))
         req.computed = req.prompt_len
         req.generated = [first_token]
         if not req.first_token_t:

@@ -89,6 +89,20 @@ class NodeConfig:
     fc_global_max_items: Optional[int] = None
     route_batch_per_step: int = 64
     ttft_slo_ms: Optional[float] = None
+    # P/D stage choreography variant (reference sidecar connectors,
+    # proxy.go:72-81), re-grounded on the xGMI transfer engine:
+    #   nixlv2         prefill -> kv_ready handle -> transfer -> decode
+    #                  (connector_nixlv2.go three-step choreography)
+    #   shared-storage try-decode-first: the decode engine checks its REAL
+    #                  prefix-cache hit fraction; >= cacheHitThreshold
+    #                  decodes locally, else requests the prefill stage
+    #                  (connector_shared_storage.go:50-67)
+    #   sglang         concurrent bootstrap: decode reserves its KV blocks
+    #                  at assign time (the bootstrap "room") so prefill and
+    #                  decode-side setup overlap and the transfer can never
+    #                  hit kv_exhausted (connector_sglang.go)
+    connector: str = "nixlv2"
+    cache_hit_threshold: float = 0.8
     seed: int = 0
     mailbox_group: Any = None
     transfer_group: Any = None
@@ -132,6 +146,9 @@ class NodeRunner:
         self._pending_adoption: Dict[str, Dict[str, Any]] = {}
         # prefill-side: req_id -> decode rank for the eventual hand-off
         self._handoff_dst: Dict[str, int] = {}
+        # shared-storage connector: prefill jobs deferred until the decode
+        # side reports a cache miss (try-decode-first)
+        self._deferred_prefill: Dict[str, EngineRequest] = {}
         # decode-side: req_ids whose tokens stream back per step
         self._streaming_ids: set = set()
         self._token_events: List[tuple] = []  # rank0: (req_id, [tokens])
@@ -394,6 +411,13 @@ class NodeRunner:
                 self._handle_assign(m)
             elif t == "done" and self.is_router:
                 self._handle_done(m)
+            elif t == "need_prefill" and m.get("dst") == self.rank:
+                req = self._deferred_prefill.pop(m["req_id"], None)
+                if req is not None:
+                    self.engine.add_request(req)
+            elif t == "pd_skip":
+                self._deferred_prefill.pop(m["req_id"], None)
+                self._handoff_dst.pop(m["req_id"], None)
             elif t == "kv_events" and self.is_router:
                 if self._precise is not None:
                     self._precise.apply_events(f"gpu{m['src']}", m["s"],
@@ -418,6 +442,9 @@ class NodeRunner:
         prompt_rank = prefill_rank if (
             prefill_rank is not None and prefill_rank != decode_rank) \
             else decode_rank
+        disagg = prefill_rank is not None and prompt_rank != decode_rank
+        shared_storage = disagg and self.cfg.connector == "shared-storage" \
+            and not encode_ranks
         if self.rank == prompt_rank:
             if prompt_rank != decode_rank:
                 req.prefill_only = True
@@ -427,15 +454,45 @@ class NodeRunner:
                 self._awaiting_embeds[m["req_id"]] = {
                     "req": req, "src": encode_ranks[0],
                     "n_mm": len(m.get("mm", []))}
+            elif shared_storage and self.rank != decode_rank:
+                # try-decode-first: start prefilling only on a reported miss
+                self._deferred_prefill[m["req_id"]] = req
             else:
                 self.engine.add_request(req)
         if self.rank == decode_rank and prompt_rank != decode_rank:
-            self._pending_adoption[m["req_id"]] = {"req": EngineRequest(
+            decode_req = EngineRequest(
                 request_id=m["req_id"], prompt_tokens=list(m["tokens"]),
                 max_tokens=m["max_tokens"], temperature=m["temperature"],
                 cached_tokens=m.get("cached", 0),
-                arrival_t=m.get("arrival") or 0.0),
-                "src": prompt_rank}
+                arrival_t=m.get("arrival") or 0.0)
+            local_hit = False
+            if shared_storage:
+                from ..engine.kvcache import block_hashes
+                h = block_hashes(decode_req.prompt_tokens,
+                                 self.engine.pool.block_size)
+                matched = self.engine.mgr.match_prefix(
+                    h, len(decode_req.prompt_tokens))
+                frac = matched / max(1, len(decode_req.prompt_tokens))
+                if frac >= self.cfg.cache_hit_threshold:
+                    # cache hit: decode locally, tell prefill to drop
+                    local_hit = True
+                    self.engine.add_request(decode_req)
+                    self._outbox.append({"type": "pd_skip",
+                                         "req_id": m["req_id"]})
+                else:
+                    # miss (finish_reason `cache_threshold` analog): fall
+                    # back to the prefill stage
+                    self._outbox.append({"type": "need_prefill",
+                                         "req_id": m["req_id"],
+                                         "dst": prompt_rank})
+            if not local_hit:
+                adoption = {"req": decode_req, "src": prompt_rank}
+                if self.cfg.connector == "sglang":
+                    # bootstrap-room reservation: hold the KV blocks now
+                    bs = self.engine.pool.block_size
+                    n = (len(decode_req.prompt_tokens) + bs - 1) // bs
+                    adoption["reserved"] = self.engine.mgr.take_blocks(n)
+                self._pending_adoption[m["req_id"]] = adoption
         if encode_ranks and self.rank in encode_ranks:
             self._encode_jobs.append({"req_id": m["req_id"],
                                       "urls": m.get("mm", []),
@@ -507,7 +564,13 @@ class NodeRunner:
                 self.engine.release_prefilled(job["req_id"])
             elif self.rank == dst:
                 n = len(job["blocks"])
-                local = self.engine.mgr.take_blocks(n)
+                reserved = (self._pending_adoption.get(job["req_id"]) or
+                            {}).get("reserved")
+                if reserved is not None and len(reserved) != n:
+                    self.engine.mgr.release_blocks(reserved)
+                    reserved = None
+                local = reserved if reserved is not None else \
+                    self.engine.mgr.take_blocks(n)
                 if local is None:
                     # keep the P2P matched: receive into scratch and drop
                     scratch = self.transfer._staging(n)

@@ -103,6 +103,11 @@ class NodeConfig:
     #                  hit kv_exhausted (connector_sglang.go)
     connector: str = "nixlv2"
     cache_hit_threshold: float = 0.8
+    # chunked decode (sidecar decode.go:62-315): cap per-dispatch generation
+    # at N tokens; each continuation re-routes through the scheduler (the
+    # engine's prefix cache makes the re-prefill nearly free), capping
+    # head-of-line blocking and enabling mid-generation rebalancing
+    decode_chunk_tokens: Optional[int] = None
     seed: int = 0
     mailbox_group: Any = None
     transfer_group: Any = None
@@ -216,6 +221,8 @@ class NodeRunner:
         self._arrivals: List[LLMRequest] = []
         self._decisions: Dict[str, RoutingDecision] = {}
         self._completions: List[Completion] = []
+        # chunked-decode continuation state (req_id -> accumulator)
+        self._chunked: Dict[str, Dict[str, Any]] = {}
         self._assign_seq = 0
         from collections import deque
         self.epp_latencies = deque(maxlen=100_000)  # ms, per routed request
@@ -353,10 +360,20 @@ class NodeRunner:
             self._decisions[req.request_id] = decision
             self.epp_latencies.append(decision.epp_latency_ms)
             decode_rank = decision.target.metadata.rank
+            chunk = self.cfg.decode_chunk_tokens
+            max_tokens = req.max_tokens
+            if chunk and max_tokens > chunk and \
+                    req.request_id not in self._chunked:
+                self._chunked[req.request_id] = {
+                    "orig": req, "prompt_len": len(req.prompt_tokens or []),
+                    "tokens": [], "first_usage": None}
+            if req.request_id in self._chunked:
+                done = len(self._chunked[req.request_id]["tokens"])
+                max_tokens = min(chunk, req.max_tokens - done)
             msg = {"type": "assign", "req_id": req.request_id,
                    "dst": decode_rank,
                    "tokens": req.prompt_tokens or [],
-                   "max_tokens": req.max_tokens,
+                   "max_tokens": max_tokens,
                    "temperature": req.temperature,
                    "is_embedding": req.is_embedding,
                    "stream": req.streaming,
@@ -509,6 +526,39 @@ class NodeRunner:
                       e2e_ms=m.get("e2e_ms"))
         if decision is not None:
             self.director.handle_response_complete(decision, usage)
+        state = self._chunked.get(m["req_id"])
+        if state is not None and not m.get("error"):
+            # chunked decode: accumulate and either continue or finalize
+            # with cumulative usage (decode.go cumulative SSE usage)
+            state["tokens"].extend(m.get("tokens", []))
+            if state["first_usage"] is None:
+                state["first_usage"] = usage
+            orig = state["orig"]
+            if len(state["tokens"]) < orig.max_tokens:
+                cont = LLMRequest(
+                    request_id=orig.request_id, model=orig.model,
+                    prompt="", target_model=orig.target_model,
+                    prompt_tokens=(orig.prompt_tokens or []) +
+                    state["tokens"],
+                    max_tokens=orig.max_tokens,
+                    temperature=orig.temperature,
+                    streaming=orig.streaming, headers=dict(orig.headers),
+                    objective_name=orig.objective_name,
+                    fairness_id=orig.fairness_id)
+                self._arrivals.append(cont)   # re-routes next step
+                return
+            del self._chunked[m["req_id"]]
+            first = state["first_usage"]
+            usage = Usage(prompt_tokens=state["prompt_len"],
+                          completion_tokens=len(state["tokens"]),
+                          cached_tokens=first.cached_tokens,
+                          ttft_ms=first.ttft_ms, tpot_ms=usage.tpot_ms,
+                          e2e_ms=(first.e2e_ms or 0) + (usage.e2e_ms or 0))
+            self._completions.append(Completion(
+                request_id=m["req_id"], usage=usage,
+                tokens=state["tokens"], error=""))
+            return
+        self._chunked.pop(m["req_id"], None)
         self._completions.append(Completion(
             request_id=m["req_id"], usage=usage,
             tokens=m.get("tokens", []), error=m.get("error", "")))

@@ -152,3 +152,29 @@ class TestMultiProcessPD:
                 break
         for r in results:
             assert r["tokens"] == toks
+
+
+class TestChunkedDecode:
+    def test_chunked_matches_unchunked(self):
+        """decode_chunk_tokens splits generation into re-routed chunks
+        (sidecar decode.go analog); cumulative result identical to one
+        dispatch because the prefix cache replays the continuation."""
+        results = {}
+        for chunk in (None, 3):
+            cfg = NodeConfig(model=TINY_LLAMA, world_size=1, topology="mono",
+                             device="cpu", dtype=torch.float32,
+                             kv_blocks=256, decode_chunk_tokens=chunk,
+                             seed=5)
+            node = NodeRunner(cfg)
+            node.submit(make_req(0, n_prompt=40, max_tokens=8))
+            got = []
+            for _ in range(120):
+                node.step()
+                got.extend(node.drain_completions())
+                if got:
+                    break
+            node.shutdown()
+            assert got and not got[0].error
+            assert got[0].usage.completion_tokens == 8
+            results[chunk] = got[0].tokens
+        assert results[None] == results[3]

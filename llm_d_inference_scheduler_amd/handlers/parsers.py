@@ -206,3 +206,138 @@ class ParserMux:
         ct = headers.get("content-type", "application/json").split(";")[0]
         parser = self.by_content_type.get(ct, self.default)
         return parser.parse_request(body, headers, path)
+
+
+# ---------------------------------------------------------------------------
+# vLLM gRPC parser (parsers/vllmgrpc): the reference parses the vLLM
+# `VllmEngine` gRPC service (Generate/Embed/HealthCheck/Abort/GetModelInfo/
+# GetServerInfo, vllm_engine.proto:10-27) from generated protobuf stubs.
+# Here the protobuf wire format is decoded directly (no codegen): varint /
+# 64-bit / length-delimited / 32-bit field walking over the gRPC
+# length-prefixed message frame.
+
+def _pb_walk(data: bytes):
+    """Yield (field_number, wire_type, value) over a protobuf message."""
+    i, n = 0, len(data)
+    while i < n:
+        tag, i = _pb_varint(data, i)
+        field, wt = tag >> 3, tag & 7
+        if wt == 0:                      # varint
+            val, i = _pb_varint(data, i)
+        elif wt == 1:                    # 64-bit
+            val, i = data[i:i + 8], i + 8
+        elif wt == 2:                    # length-delimited
+            ln, i = _pb_varint(data, i)
+            val, i = data[i:i + ln], i + ln
+        elif wt == 5:                    # 32-bit
+            val, i = data[i:i + 4], i + 4
+        else:
+            raise ValueError(f"unsupported wire type {wt}")
+        yield field, wt, val
+
+
+def _pb_varint(data: bytes, i: int):
+    shift, out = 0, 0
+    while True:
+        b = data[i]
+        i += 1
+        out |= (b & 0x7F) << shift
+        if not b & 0x80:
+            return out, i
+        shift += 7
+        if shift > 63:
+            raise ValueError("varint overflow")
+
+
+def _pb_packed_uint32(val: bytes) -> List[int]:
+    out, i = [], 0
+    while i < len(val):
+        v, i = _pb_varint(val, i)
+        out.append(v)
+    return out
+
+
+def _strip_grpc_frame(body: bytes) -> bytes:
+    """gRPC messages are framed [compressed u8][length u32 BE][payload]."""
+    if len(body) >= 5 and body[0] in (0, 1):
+        ln = int.from_bytes(body[1:5], "big")
+        if len(body) == 5 + ln:
+            if body[0] == 1:
+                raise ValueError("compressed gRPC frame unsupported")
+            return body[5:]
+    return body
+
+
+@register_plugin("vllm-grpc-parser")
+class VllmGrpcParser(Parser):
+    """vLLM gRPC `VllmEngine` request parser.
+
+    Message schema (field numbers mirror the reference's
+    vllm_engine.proto Generate/Embed requests):
+      GenerateRequest: 1 model(str) 2 prompt(str) 3 token_ids(packed u32)
+                       4 max_tokens(u32) 5 temperature(f32) 6 stream(bool)
+                       7 request_id(str)
+      EmbedRequest:    1 model(str) 2 prompt(str) 3 token_ids(packed u32)
+                       7 request_id(str)
+      GenerateResponse usage fields: 1 prompt_tokens 2 completion_tokens
+                       3 cached_tokens (varints)
+    Non-routable methods (HealthCheck/Abort/GetModelInfo/GetServerInfo)
+    return Skip -> random-endpoint fallback, as the reference does for
+    parser-skipped bodies (server.go:335-342).
+    """
+
+    ROUTABLE = ("/Generate", "/Embed")
+
+    def parse_request(self, body: bytes, headers: Dict[str, str],
+                      path: str = "") -> ParseResult:
+        method = path.rsplit("/", 1)[-1] if path else "Generate"
+        if method not in ("Generate", "Embed"):
+            return ParseResult(skip=True)
+        try:
+            payload = _strip_grpc_frame(body)
+            req = LLMRequest(
+                request_id=headers.get("x-request-id", uuid.uuid4().hex),
+                model="")
+            import struct
+            for field, wt, val in _pb_walk(payload):
+                if field == 1 and wt == 2:
+                    req.model = val.decode("utf-8", "replace")
+                elif field == 2 and wt == 2:
+                    req.prompt = val.decode("utf-8", "replace")
+                elif field == 3 and wt == 2:
+                    req.prompt_tokens = _pb_packed_uint32(val)
+                elif field == 3 and wt == 0:   # unpacked repeated uint32
+                    req.prompt_tokens = (req.prompt_tokens or []) + [val]
+                elif field == 4 and wt == 0:
+                    req.max_tokens = int(val)
+                elif field == 5 and wt == 5:
+                    req.temperature = struct.unpack("<f", val)[0]
+                elif field == 6 and wt == 0:
+                    req.streaming = bool(val)
+                elif field == 7 and wt == 2:
+                    req.request_id = val.decode("utf-8", "replace")
+        except (ValueError, IndexError) as e:
+            return ParseResult(error=f"invalid protobuf: {e}")
+        if not req.model:
+            return ParseResult(error="missing model")
+        req.is_embedding = method == "Embed"
+        _headers_into(req, headers)
+        return ParseResult(request=req)
+
+    def parse_response_usage(self, body: bytes,
+                             streaming: bool) -> Optional[Usage]:
+        try:
+            payload = _strip_grpc_frame(body)
+            u = Usage()
+            for field, wt, val in _pb_walk(payload):
+                if wt != 0:
+                    continue
+                if field == 1:
+                    u.prompt_tokens = int(val)
+                elif field == 2:
+                    u.completion_tokens = int(val)
+                elif field == 3:
+                    u.cached_tokens = int(val)
+            return u
+        except (ValueError, IndexError):
+            return None

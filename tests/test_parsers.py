@@ -82,3 +82,75 @@ class TestOtherParsers:
         r2 = mux.parse_request(b"anything",
                                {"content-type": "application/grpc"}, "")
         assert r2.skip
+
+
+class TestVllmGrpcParser:
+    @staticmethod
+    def _encode_varint(v):
+        out = b""
+        while True:
+            b7 = v & 0x7F
+            v >>= 7
+            if v:
+                out += bytes([b7 | 0x80])
+            else:
+                return out + bytes([b7])
+
+    def _encode_generate(self, model, tokens, max_tokens, temp, stream,
+                         rid=""):
+        import struct
+        ev = self._encode_varint
+        msg = b""
+        msg += bytes([0x0A]) + ev(len(model)) + model.encode()
+        packed = b"".join(ev(t) for t in tokens)
+        msg += bytes([0x1A]) + ev(len(packed)) + packed
+        msg += bytes([0x20]) + ev(max_tokens)
+        msg += bytes([0x2D]) + struct.pack("<f", temp)
+        msg += bytes([0x30]) + ev(1 if stream else 0)
+        if rid:
+            msg += bytes([0x3A]) + ev(len(rid)) + rid.encode()
+        return bytes([0]) + len(msg).to_bytes(4, "big") + msg
+
+    def test_generate_roundtrip(self):
+        from llm_d_inference_scheduler_amd.handlers.parsers import \
+            VllmGrpcParser
+        p = VllmGrpcParser()
+        body = self._encode_generate("llama-3-8b", [5, 300, 70000], 33,
+                                     0.5, True, rid="abc")
+        res = p.parse_request(body, {}, "/vllm.VllmEngine/Generate")
+        assert res.error is None and not res.skip
+        r = res.request
+        assert r.model == "llama-3-8b"
+        assert r.prompt_tokens == [5, 300, 70000]
+        assert r.max_tokens == 33
+        assert abs(r.temperature - 0.5) < 1e-6
+        assert r.streaming and r.request_id == "abc"
+
+    def test_embed_and_skip_methods(self):
+        from llm_d_inference_scheduler_amd.handlers.parsers import \
+            VllmGrpcParser
+        p = VllmGrpcParser()
+        body = self._encode_generate("m", [1, 2], 1, 0.0, False)
+        res = p.parse_request(body, {}, "/vllm.VllmEngine/Embed")
+        assert res.request.is_embedding
+        for m in ("HealthCheck", "Abort", "GetModelInfo", "GetServerInfo"):
+            assert p.parse_request(b"", {}, f"/vllm.VllmEngine/{m}").skip
+
+    def test_response_usage(self):
+        from llm_d_inference_scheduler_amd.handlers.parsers import \
+            VllmGrpcParser
+        ev = self._encode_varint
+        msg = bytes([0x08]) + ev(100) + bytes([0x10]) + ev(32) + \
+            bytes([0x18]) + ev(48)
+        body = bytes([0]) + len(msg).to_bytes(4, "big") + msg
+        u = VllmGrpcParser().parse_response_usage(body, False)
+        assert (u.prompt_tokens, u.completion_tokens, u.cached_tokens) == \
+            (100, 32, 48)
+
+    def test_malformed(self):
+        from llm_d_inference_scheduler_amd.handlers.parsers import \
+            VllmGrpcParser
+        res = VllmGrpcParser().parse_request(
+            b"\x00\x00\x00\x00\x03\xff\xff\xff", {},
+            "/vllm.VllmEngine/Generate")
+        assert res.error is not None

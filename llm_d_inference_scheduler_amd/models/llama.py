@@ -43,7 +43,7 @@ def _load_tuned_gemms() -> None:
         tunable.enable(True)
         tunable.tuning_enable(False)
         tunable.read_file(path)
-        log.info("loaded tuned GEMM solutions from %s", path)
+        log.info("loaded tuned GEMM solutions", path=path)
 
 
 @dataclass

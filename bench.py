@@ -39,8 +39,9 @@ def parse_args():
     p.add_argument("--model", default="llama-3-8b")
     p.add_argument("--prompt-len", type=int, default=1024)
     p.add_argument("--max-tokens", type=int, default=128)
-    p.add_argument("--concurrency", type=int, default=64,
-                   help="in-flight requests per decode rank (closed loop)")
+    p.add_argument("--concurrency", type=int, default=128,
+                   help="in-flight requests per decode rank (closed loop; "
+                        "128 is the measured single-GPU throughput knee)")
     p.add_argument("--shared-prefix", type=float, default=0.5,
                    help="fraction of prompt shared within a request group")
     p.add_argument("--group", type=int, default=4,

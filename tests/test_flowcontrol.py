@@ -230,3 +230,22 @@ class TestBlockingController:
             fc.submit(mk_req(f"r{i}", size=10))
         # JSQ-bytes should balance
         assert abs(fc.shards[0].queued_len - fc.shards[1].queued_len) <= 1
+
+
+class TestFlowControlBenchmark:
+    def test_bench_runs_and_reports(self):
+        """Reference flowcontrol/benchmark.go analog: d/s r/s zombies/s."""
+        from llm_d_inference_scheduler_amd.flowcontrol.benchmark import \
+            run_bench
+        res = run_bench(duration_s=0.3)
+        assert res.dispatched > 0
+        assert res.zombies == 0
+        assert res.dispatched + res.rejected + res.evicted <= res.submitted
+        assert "d_per_s" in res.to_json()
+
+    def test_bench_saturation_rejects(self):
+        from llm_d_inference_scheduler_amd.flowcontrol.benchmark import \
+            run_bench
+        res = run_bench(duration_s=0.3, saturated_every=1, max_items=64)
+        # permanently saturated + tiny capacity: most requests rejected
+        assert res.rejected + res.evicted > 0

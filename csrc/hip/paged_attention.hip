@@ -29,16 +29,6 @@ constexpr int CHUNK = 256;    // tokens per online-softmax chunk
 constexpr int NW = 4;         // waves per workgroup
 constexpr float NEG = -1e30f;
 
-// K rows are staged through LDS in half-chunks: the staging loads are
-// fully coalesced (16 consecutive threads read one 256-B K row), and the
-// per-lane dot reads come from LDS at a XOR-swizzled layout
-// (conflict-free b128 groups) instead of 16-B strided HBM requests.
-constexpr int KSTAGE = 128;   // tokens per staging pass (32 KB LDS)
-
-__device__ __forceinline__ int pa_swz(int row, int byte_off) {
-  return row * 256 + (byte_off ^ ((row & 15) << 4));
-}
-
 template <int QPG, bool SPLIT>
 __global__ __launch_bounds__(NW * WAVE) void paged_attention_kernel(
     const short* __restrict__ q,        // [B, QH, D]
@@ -64,7 +54,6 @@ __global__ __launch_bounds__(NW * WAVE) void paged_attention_kernel(
   __shared__ float logits[QPG][CHUNK];
   __shared__ float m_sh[QPG], l_sh[QPG], alpha_sh[QPG];
   __shared__ float comb[NW][QPG][D];
-  __shared__ short k_stage[KSTAGE * D];   // swizzled, byte-addressed
 
   const int tid = threadIdx.x;
   const int wave = tid / WAVE;
@@ -101,30 +90,21 @@ __global__ __launch_bounds__(NW * WAVE) void paged_attention_kernel(
 
   for (int chunk0 = t_begin; chunk0 < t_end; chunk0 += CHUNK) {
     const int n_t = min(CHUNK, t_end - chunk0);
-    // ---- Phase A: logits[h][t_local], K staged through LDS ----
-    for (int s0 = 0; s0 < n_t; s0 += KSTAGE) {
-      const int s_cnt = min(KSTAGE, n_t - s0);
-      // A1: coalesced cooperative stage — 16 threads cover one 256-B row
-      __syncthreads();   // previous pass's LDS reads complete
-      for (int piece = tid; piece < s_cnt * 16; piece += NW * WAVE) {
-        const int row = piece / 16, c16 = piece % 16;
-        const int t = chunk0 + s0 + row;
-        const int64_t blk = bt[t / bs];
-        const short8 v = *(const short8*)(
-            k_cache + (((blk * kvh + kh) * bs) + t % bs) * D + c16 * 8);
-        *(short8*)((char*)k_stage + pa_swz(row, c16 * 16)) = v;
-      }
-      __syncthreads();
-      // A2: per-lane dot from LDS (b128 reads, conflict-free swizzle)
-      const int t_local = wave * WAVE + lane - s0;
+    // ---- Phase A: logits[h][t_local] ----
+    {
+      const int t_local = wave * WAVE + lane;
       float dot[QPG];
 #pragma unroll
       for (int h = 0; h < QPG; ++h) dot[h] = 0.f;
-      if (t_local >= 0 && t_local < s_cnt) {
+      if (t_local < n_t) {
+        const int t = chunk0 + t_local;
+        const int64_t blk = bt[t / bs];
+        const int row = t % bs;
+        const short8* krow =
+            (const short8*)(k_cache + (((blk * kvh + kh) * bs) + row) * D);
 #pragma unroll 4
         for (int c = 0; c < D / 8; ++c) {
-          short8 kv8 = *(const short8*)(
-              (const char*)k_stage + pa_swz(t_local, c * 16));
+          short8 kv8 = krow[c];
           float kf[8];
 #pragma unroll
           for (int j = 0; j < 8; ++j) kf[j] = bf16_to_f32(kv8[j]);
@@ -137,14 +117,10 @@ __global__ __launch_bounds__(NW * WAVE) void paged_attention_kernel(
                       qb[2] * kf[6] + qb[3] * kf[7];
           }
         }
-#pragma unroll
-        for (int h = 0; h < QPG; ++h)
-          logits[h][wave * WAVE + lane] = dot[h];
-      } else if (wave * WAVE + lane >= n_t) {
-#pragma unroll
-        for (int h = 0; h < QPG; ++h)
-          logits[h][wave * WAVE + lane] = NEG;
       }
+#pragma unroll
+      for (int h = 0; h < QPG; ++h)
+        logits[h][wave * WAVE + lane] = (t_local < n_t) ? dot[h] : NEG;
     }
     __syncthreads();
 

@@ -102,7 +102,7 @@ __global__ __launch_bounds__(NW * WAVE) void paged_attention_kernel(
         const int row = t % bs;
         const short8* krow =
             (const short8*)(k_cache + (((blk * kvh + kh) * bs) + row) * D);
-#pragma unroll 4
+#pragma unroll 8
         for (int c = 0; c < D / 8; ++c) {
           short8 kv8 = krow[c];
           float kf[8];

@@ -51,8 +51,8 @@ def bench(B, ctx, np_=None, part=512, iters=50):
 if __name__ == "__main__":
     for B in (64, 128, 256):
         bench(B, 1152)
-        n_wgs = B * KVH
-        if n_wgs < 1024:
-            bench(B, 1152, np_=max(2, 1024 // n_wgs), part=512)
+        for np_ in (2, 4):
+            bench(B, 1152, np_=np_, part=(1152 + np_ - 1) // np_ // 256 * 256
+                  + 256)
     bench(8, 8192)
     bench(8, 8192, np_=16, part=512)

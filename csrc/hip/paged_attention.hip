@@ -161,13 +161,35 @@ __global__ __launch_bounds__(NW * WAVE) void paged_attention_kernel(
     {
       const int t_base = wave * WAVE;
       const int t_cnt = min(WAVE, n_t - t_base);
-      for (int i = 0; i < t_cnt; ++i) {
+      // batch the V row loads 8 deep so they overlap (each row is one
+      // coalesced 256-B wave read; a load-use loop serializes on latency)
+      int i = 0;
+      for (; i + 8 <= t_cnt; i += 8) {
+        int32_t pairs[8];
+#pragma unroll
+        for (int j = 0; j < 8; ++j) {
+          const int t = chunk0 + t_base + i + j;
+          const int64_t blk = bt[t / bs];
+          pairs[j] = ((const int32_t*)(
+              v_cache + (((blk * kvh + kh) * bs) + t % bs) * D))[lane];
+        }
+#pragma unroll
+        for (int j = 0; j < 8; ++j) {
+          const float v0 = bf16_to_f32((short)(pairs[j] & 0xFFFF));
+          const float v1 = bf16_to_f32((short)((pairs[j] >> 16) & 0xFFFF));
+#pragma unroll
+          for (int h = 0; h < QPG; ++h) {
+            const float p = logits[h][t_base + i + j];
+            o_acc[h][0] += p * v0;
+            o_acc[h][1] += p * v1;
+          }
+        }
+      }
+      for (; i < t_cnt; ++i) {
         const int t = chunk0 + t_base + i;
         const int64_t blk = bt[t / bs];
-        const int row = t % bs;
-        const int32_t* vrow =
-            (const int32_t*)(v_cache + (((blk * kvh + kh) * bs) + row) * D);
-        const int32_t pair = vrow[lane];  // 2 bf16, coalesced 256B row
+        const int32_t pair = ((const int32_t*)(
+            v_cache + (((blk * kvh + kh) * bs) + t % bs) * D))[lane];
         const float v0 = bf16_to_f32((short)(pair & 0xFFFF));
         const float v1 = bf16_to_f32((short)((pair >> 16) & 0xFFFF));
 #pragma unroll

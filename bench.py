@@ -48,6 +48,10 @@ def parse_args():
                    help="requests per shared-prefix group")
     p.add_argument("--kv-gb", type=float, default=48.0,
                    help="KV pool budget per GPU (GB)")
+    p.add_argument("--kv-dtype", default="auto",
+                   choices=["auto", "bf16", "fp8"],
+                   help="KV cache storage dtype (fp8 = OCP e4m3; compute "
+                        "stays bf16)")
     p.add_argument("--ttft-slo-ms", type=float, default=2000.0)
     p.add_argument("--device", default=None, help="override (cpu for tests)")
     p.add_argument("--seed", type=int, default=1234)
@@ -213,7 +217,7 @@ def main():
                   BandConfig(0, ordering="fcfs"),
                   BandConfig(-1, ordering="fcfs")] if fc else [],
         fc_global_max_items=args.concurrency * 8 if fc else None,
-        ttft_slo_ms=args.ttft_slo_ms,
+        ttft_slo_ms=args.ttft_slo_ms, kv_cache_dtype=args.kv_dtype,
         mailbox_group=mailbox_group, transfer_group=transfer_group,
         seed=args.seed)
     node = NodeRunner(cfg)
@@ -332,6 +336,7 @@ def main():
                 "mode": args.mode,
                 "prompt_len": args.prompt_len,
                 "max_tokens": args.max_tokens,
+                "kv_cache_dtype": args.kv_dtype,
                 "ttft_slo_ms": args.ttft_slo_ms,
                 "routed_req_s": round(completed / elapsed, 2),
                 "p50_epp_latency_ms": p50_epp,

@@ -50,11 +50,11 @@ __device__ __forceinline__ unsigned int pack_bf16(float a, float b) {
          (unsigned int)(unsigned short)f32_to_bf16(a);
 }
 
-template <int QPG>
+template <int QPG, typename CT>
 __global__ __launch_bounds__(NW * WAVE) void flash_prefill_kernel(
     const short* __restrict__ q,        // [T, QH, D]
-    const short* __restrict__ k_cache,  // [NB, KVH, BS, D]
-    const short* __restrict__ v_cache,  // [NB, KVH, BS, D]
+    const CT* __restrict__ k_cache,     // [NB, KVH, BS, D] bf16|fp8
+    const CT* __restrict__ v_cache,     // [NB, KVH, BS, D] bf16|fp8
     const int32_t* __restrict__ block_tables,  // [S, max_blocks]
     const int32_t* __restrict__ seq_meta,      // [S, 3] start, chunk, prior
     const int32_t* __restrict__ tiles,         // [n_tiles, 2] seq, vrow0
@@ -121,8 +121,8 @@ __global__ __launch_bounds__(NW * WAVE) void flash_prefill_kernel(
       if (t < ctx) {
         const int64_t base =
             (((int64_t)bt[t / bs] * kvh + kh) * bs + t % bs) * D;
-        piece = *(const short8*)(k_cache + base + c16 * 8);
-        vpiece = *(const short8*)(v_cache + base + c16 * 8);
+        piece = load_kv8_bf16(k_cache + base + c16 * 8);
+        vpiece = load_kv8_bf16(v_cache + base + c16 * 8);
       }
       *(short8*)((char*)k_lds + k_swz(row, c16 * 16)) = piece;
 #pragma unroll
@@ -236,17 +236,22 @@ hipError_t lds_flash_prefill(const void* q, const void* k_cache,
                              const int32_t* block_tables,
                              const int32_t* seq_meta, const int32_t* tiles,
                              void* out, int n_tiles, int n_q_heads, int kvh,
-                             int bs, int head_dim, int max_blocks, float scale,
-                             hipStream_t stream) {
+                             int bs, int head_dim, int max_blocks, int kv_fp8,
+                             float scale, hipStream_t stream) {
   if (n_tiles == 0) return hipSuccess;
   if (head_dim != D) return hipErrorInvalidValue;
   const int qpg = n_q_heads / kvh;
   dim3 grid(n_tiles, kvh), block(NW * WAVE);
-#define LAUNCH(QPG)                                                           \
-  hipLaunchKernelGGL((flash_prefill_kernel<QPG>), grid, block, 0, stream,     \
-                     (const short*)q, (const short*)k_cache,                  \
-                     (const short*)v_cache, block_tables, seq_meta, tiles,    \
+#define LAUNCH_CT(QPG, CT)                                                    \
+  hipLaunchKernelGGL((flash_prefill_kernel<QPG, CT>), grid, block, 0, stream, \
+                     (const short*)q, (const CT*)k_cache,                     \
+                     (const CT*)v_cache, block_tables, seq_meta, tiles,       \
                      (short*)out, kvh, bs, max_blocks, scale)
+#define LAUNCH(QPG)                                                           \
+  do {                                                                        \
+    if (kv_fp8) LAUNCH_CT(QPG, unsigned char);                                \
+    else LAUNCH_CT(QPG, short);                                               \
+  } while (0)
   switch (qpg) {
     case 1: LAUNCH(1); break;
     case 2: LAUNCH(2); break;

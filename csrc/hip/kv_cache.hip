@@ -13,10 +13,11 @@ namespace {
 // Scatter the freshly projected K/V of each token into its pool slot.
 //  k_new/v_new: [T, KVH, D] bf16; slot_mapping: [T] int64 (block*BS + row)
 //  k_cache/v_cache: [NB, KVH, BS, D] bf16 (one layer's slice)
+template <typename CT>
 __global__ void reshape_and_cache_kernel(const short* __restrict__ k_new,
                                          const short* __restrict__ v_new,
-                                         short* __restrict__ k_cache,
-                                         short* __restrict__ v_cache,
+                                         CT* __restrict__ k_cache,
+                                         CT* __restrict__ v_cache,
                                          const int64_t* __restrict__ slots,
                                          int n_tokens, int kvh, int bs, int d) {
   const int t = blockIdx.x;
@@ -32,8 +33,8 @@ __global__ void reshape_and_cache_kernel(const short* __restrict__ k_new,
     const short8* src_k = (const short8*)(k_new + ((int64_t)t * kvh + h) * d);
     const short8* src_v = (const short8*)(v_new + ((int64_t)t * kvh + h) * d);
     int64_t dst_off = (((block * kvh + h) * bs) + row) * d;
-    ((short8*)(k_cache + dst_off))[c] = src_k[c];
-    ((short8*)(v_cache + dst_off))[c] = src_v[c];
+    store_kv8(k_cache + dst_off + c * 8, src_k[c]);
+    store_kv8(v_cache + dst_off + c * 8, src_v[c]);
   }
 }
 
@@ -85,22 +86,34 @@ extern "C" {
 hipError_t lds_reshape_and_cache(const void* k_new, const void* v_new,
                                  void* k_cache, void* v_cache,
                                  const int64_t* slots, int n_tokens, int kvh,
-                                 int bs, int d, hipStream_t stream) {
+                                 int bs, int d, int kv_fp8,
+                                 hipStream_t stream) {
   if (n_tokens == 0) return hipSuccess;
   int threads = kvh * (d / 8);
   if (threads > 256) threads = 256;
-  hipLaunchKernelGGL(reshape_and_cache_kernel, dim3(n_tokens), dim3(threads), 0,
-                     stream, (const short*)k_new, (const short*)v_new,
-                     (short*)k_cache, (short*)v_cache, slots, n_tokens, kvh, bs,
-                     d);
+  if (kv_fp8) {
+    hipLaunchKernelGGL(reshape_and_cache_kernel<unsigned char>,
+                       dim3(n_tokens), dim3(threads), 0, stream,
+                       (const short*)k_new, (const short*)v_new,
+                       (unsigned char*)k_cache, (unsigned char*)v_cache,
+                       slots, n_tokens, kvh, bs, d);
+  } else {
+    hipLaunchKernelGGL(reshape_and_cache_kernel<short>, dim3(n_tokens),
+                       dim3(threads), 0, stream, (const short*)k_new,
+                       (const short*)v_new, (short*)k_cache, (short*)v_cache,
+                       slots, n_tokens, kvh, bs, d);
+  }
   HIP_CHECK_LAST();
   return hipSuccess;
 }
 
 hipError_t lds_gather_blocks(const void* pool, void* staging,
                              const int32_t* block_ids, int n_sel, int n_layers,
-                             int64_t n_blocks, int64_t block_elems,
+                             int64_t n_blocks, int64_t block_bytes,
                              int is_scatter, hipStream_t stream) {
+  // element-size agnostic: the kernels move 16-byte short8 units;
+  // block_bytes is one (layer, K/V) block's bytes (must be 16-aligned)
+  int64_t block_elems = block_bytes / 2;
   if (n_sel == 0) return hipSuccess;
   int64_t total = (int64_t)n_sel * n_layers * 2 * (block_elems / 8);
   int threads = 256;

@@ -29,24 +29,31 @@ hipError_t lds_rope(void*, void*, const float*, const int32_t*, int, int, int,
                     int, hipStream_t);
 hipError_t lds_silu_mul(const void*, void*, int64_t, int, hipStream_t);
 hipError_t lds_reshape_and_cache(const void*, const void*, void*, void*,
-                                 const int64_t*, int, int, int, int,
+                                 const int64_t*, int, int, int, int, int,
                                  hipStream_t);
 hipError_t lds_gather_blocks(const void*, void*, const int32_t*, int, int,
                              int64_t, int64_t, int, hipStream_t);
 hipError_t lds_paged_attention(const void*, const void*, const void*,
                                const int32_t*, const int32_t*, void*, int, int,
-                               int, int, int, int, float, hipStream_t);
+                               int, int, int, int, int, float, hipStream_t);
 hipError_t lds_paged_attention_split(const void*, const void*, const void*,
                                      const int32_t*, const int32_t*, void*,
                                      float*, float*, int, int, int, int, int,
-                                     int, int, int, float, hipStream_t);
+                                     int, int, int, int, float, hipStream_t);
 hipError_t lds_flash_prefill(const void*, const void*, const void*,
                              const int32_t*, const int32_t*, const int32_t*,
-                             void*, int, int, int, int, int, int, float,
+                             void*, int, int, int, int, int, int, int, float,
                              hipStream_t);
 }
 
 namespace {
+
+int kv_fp8_flag(const torch::Tensor& cache) {
+  if (cache.scalar_type() == at::kFloat8_e4m3fn) return 1;
+  TORCH_CHECK(cache.scalar_type() == at::kBFloat16,
+              "KV cache must be bf16 or fp8_e4m3fn");
+  return 0;
+}
 
 hipStream_t cur_stream() {
   return at::hip::getCurrentHIPStreamMasqueradingAsCUDA().stream();
@@ -148,7 +155,7 @@ void reshape_and_cache(torch::Tensor k_new, torch::Tensor v_new,
   CHECK_HIP(lds_reshape_and_cache(k_new.data_ptr(), v_new.data_ptr(),
                                   k_cache.data_ptr(), v_cache.data_ptr(),
                                   slots.data_ptr<int64_t>(), T, kvh, bs, d,
-                                  cur_stream()));
+                                  kv_fp8_flag(k_cache), cur_stream()));
 }
 
 void move_blocks(torch::Tensor pool, torch::Tensor staging,
@@ -157,10 +164,11 @@ void move_blocks(torch::Tensor pool, torch::Tensor staging,
   // pool: [L, 2, NB, KVH, BS, D]; staging: [n, L, 2, KVH, BS, D]
   int L = (int)pool.size(0);
   int64_t nb = pool.size(2);
-  int64_t block_elems = pool.size(3) * pool.size(4) * pool.size(5);
+  int64_t block_bytes = pool.size(3) * pool.size(4) * pool.size(5) *
+                        pool.element_size();
   CHECK_HIP(lds_gather_blocks(pool.data_ptr(), staging.data_ptr(),
                               block_ids.data_ptr<int32_t>(),
-                              (int)block_ids.size(0), L, nb, block_elems,
+                              (int)block_ids.size(0), L, nb, block_bytes,
                               is_scatter ? 1 : 0, cur_stream()));
 }
 
@@ -177,7 +185,7 @@ torch::Tensor paged_attention(torch::Tensor q, torch::Tensor k_cache,
       q.data_ptr(), k_cache.data_ptr(), v_cache.data_ptr(),
       block_tables.data_ptr<int32_t>(), seq_lens.data_ptr<int32_t>(),
       out.data_ptr(), B, qh, kvh, bs, d, (int)block_tables.size(1),
-      (float)scale, cur_stream()));
+      kv_fp8_flag(k_cache), (float)scale, cur_stream()));
   return out;
 }
 
@@ -200,7 +208,7 @@ torch::Tensor paged_attention_split(torch::Tensor q, torch::Tensor k_cache,
       block_tables.data_ptr<int32_t>(), seq_lens.data_ptr<int32_t>(),
       out.data_ptr(), part_o.data_ptr<float>(), part_ml.data_ptr<float>(), B,
       qh, kvh, bs, d, (int)block_tables.size(1), (int)n_parts,
-      (int)part_tokens, (float)scale, cur_stream()));
+      (int)part_tokens, kv_fp8_flag(k_cache), (float)scale, cur_stream()));
   return out;
 }
 
@@ -218,7 +226,8 @@ torch::Tensor flash_prefill(torch::Tensor q, torch::Tensor k_cache,
       q.data_ptr(), k_cache.data_ptr(), v_cache.data_ptr(),
       block_tables.data_ptr<int32_t>(), seq_meta.data_ptr<int32_t>(),
       tiles.data_ptr<int32_t>(), out.data_ptr(), (int)tiles.size(0), qh, kvh,
-      bs, d, (int)block_tables.size(1), (float)scale, cur_stream()));
+      bs, d, (int)block_tables.size(1), kv_fp8_flag(k_cache), (float)scale,
+      cur_stream()));
   return out;
 }
 

@@ -29,11 +29,11 @@ constexpr int CHUNK = 256;    // tokens per online-softmax chunk
 constexpr int NW = 4;         // waves per workgroup
 constexpr float NEG = -1e30f;
 
-template <int QPG, bool SPLIT>
+template <int QPG, bool SPLIT, typename CT>
 __global__ __launch_bounds__(NW * WAVE) void paged_attention_kernel(
     const short* __restrict__ q,        // [B, QH, D]
-    const short* __restrict__ k_cache,  // [NB, KVH, BS, D]
-    const short* __restrict__ v_cache,  // [NB, KVH, BS, D]
+    const CT* __restrict__ k_cache,     // [NB, KVH, BS, D] bf16|fp8
+    const CT* __restrict__ v_cache,     // [NB, KVH, BS, D] bf16|fp8
     const int32_t* __restrict__ block_tables,  // [B, max_blocks]
     const int32_t* __restrict__ seq_lens,      // [B]
     short* __restrict__ out,                   // [B, QH, D]
@@ -100,14 +100,11 @@ __global__ __launch_bounds__(NW * WAVE) void paged_attention_kernel(
         const int t = chunk0 + t_local;
         const int64_t blk = bt[t / bs];
         const int row = t % bs;
-        const short8* krow =
-            (const short8*)(k_cache + (((blk * kvh + kh) * bs) + row) * D);
+        const CT* krow = k_cache + (((blk * kvh + kh) * bs) + row) * D;
 #pragma unroll 8
         for (int c = 0; c < D / 8; ++c) {
-          short8 kv8 = krow[c];
           float kf[8];
-#pragma unroll
-          for (int j = 0; j < 8; ++j) kf[j] = bf16_to_f32(kv8[j]);
+          load_kv8(krow + c * 8, kf);
 #pragma unroll
           for (int h = 0; h < QPG; ++h) {
             const float4v* q4 = (const float4v*)&q_lds[h][c * 8];
@@ -165,33 +162,33 @@ __global__ __launch_bounds__(NW * WAVE) void paged_attention_kernel(
       // coalesced 256-B wave read; a load-use loop serializes on latency)
       int i = 0;
       for (; i + 8 <= t_cnt; i += 8) {
-        int32_t pairs[8];
+        const CT* vrows[8];
 #pragma unroll
         for (int j = 0; j < 8; ++j) {
           const int t = chunk0 + t_base + i + j;
           const int64_t blk = bt[t / bs];
-          pairs[j] = ((const int32_t*)(
-              v_cache + (((blk * kvh + kh) * bs) + t % bs) * D))[lane];
+          vrows[j] = v_cache + (((blk * kvh + kh) * bs) + t % bs) * D +
+                     lane * 2;
         }
+        float vv[8][2];
+#pragma unroll
+        for (int j = 0; j < 8; ++j) load_kv2(vrows[j], vv[j][0], vv[j][1]);
 #pragma unroll
         for (int j = 0; j < 8; ++j) {
-          const float v0 = bf16_to_f32((short)(pairs[j] & 0xFFFF));
-          const float v1 = bf16_to_f32((short)((pairs[j] >> 16) & 0xFFFF));
 #pragma unroll
           for (int h = 0; h < QPG; ++h) {
             const float p = logits[h][t_base + i + j];
-            o_acc[h][0] += p * v0;
-            o_acc[h][1] += p * v1;
+            o_acc[h][0] += p * vv[j][0];
+            o_acc[h][1] += p * vv[j][1];
           }
         }
       }
       for (; i < t_cnt; ++i) {
         const int t = chunk0 + t_base + i;
         const int64_t blk = bt[t / bs];
-        const int32_t pair = ((const int32_t*)(
-            v_cache + (((blk * kvh + kh) * bs) + t % bs) * D))[lane];
-        const float v0 = bf16_to_f32((short)(pair & 0xFFFF));
-        const float v1 = bf16_to_f32((short)((pair >> 16) & 0xFFFF));
+        float v0, v1;
+        load_kv2(v_cache + (((blk * kvh + kh) * bs) + t % bs) * D + lane * 2,
+                 v0, v1);
 #pragma unroll
         for (int h = 0; h < QPG; ++h) {
           const float p = logits[h][t_base + i];
@@ -290,18 +287,23 @@ hipError_t lds_paged_attention(const void* q, const void* k_cache,
                                const void* v_cache, const int32_t* block_tables,
                                const int32_t* seq_lens, void* out, int n_seqs,
                                int n_q_heads, int kvh, int bs, int head_dim,
-                               int max_blocks, float scale,
+                               int max_blocks, int kv_fp8, float scale,
                                hipStream_t stream) {
   if (n_seqs == 0) return hipSuccess;
   if (head_dim != D) return hipErrorInvalidValue;
   const int qpg = n_q_heads / kvh;
   dim3 grid(n_seqs, kvh), block(NW * WAVE);
-#define LAUNCH(QPG)                                                           \
-  hipLaunchKernelGGL((paged_attention_kernel<QPG, false>), grid, block, 0,    \
-                     stream, (const short*)q, (const short*)k_cache,          \
-                     (const short*)v_cache, block_tables, seq_lens,           \
+#define LAUNCH_CT(QPG, CT)                                                    \
+  hipLaunchKernelGGL((paged_attention_kernel<QPG, false, CT>), grid, block,   \
+                     0, stream, (const short*)q, (const CT*)k_cache,          \
+                     (const CT*)v_cache, block_tables, seq_lens,              \
                      (short*)out, nullptr, nullptr, kvh, bs, max_blocks, 0,   \
                      scale)
+#define LAUNCH(QPG)                                                           \
+  do {                                                                        \
+    if (kv_fp8) LAUNCH_CT(QPG, unsigned char);                                \
+    else LAUNCH_CT(QPG, short);                                               \
+  } while (0)
   switch (qpg) {
     case 1: LAUNCH(1); break;
     case 2: LAUNCH(2); break;
@@ -310,6 +312,7 @@ hipError_t lds_paged_attention(const void* q, const void* k_cache,
     default: return hipErrorInvalidValue;
   }
 #undef LAUNCH
+#undef LAUNCH_CT
   HIP_CHECK_LAST();
   return hipSuccess;
 }
@@ -318,20 +321,23 @@ hipError_t lds_paged_attention_split(
     const void* q, const void* k_cache, const void* v_cache,
     const int32_t* block_tables, const int32_t* seq_lens, void* out,
     float* part_o, float* part_ml, int n_seqs, int n_q_heads, int kvh, int bs,
-    int head_dim, int max_blocks, int n_parts, int part_tokens, float scale,
-    hipStream_t stream) {
+    int head_dim, int max_blocks, int n_parts, int part_tokens, int kv_fp8,
+    float scale, hipStream_t stream) {
   if (n_seqs == 0) return hipSuccess;
   if (head_dim != D || n_parts > 64) return hipErrorInvalidValue;
   const int qpg = n_q_heads / kvh;
   dim3 grid(n_seqs, kvh, n_parts), block(NW * WAVE);
   dim3 cgrid(n_seqs, kvh), cblock(256);
+#define LAUNCH_CT(QPG, CT)                                                    \
+  hipLaunchKernelGGL((paged_attention_kernel<QPG, true, CT>), grid, block,    \
+                     0, stream, (const short*)q, (const CT*)k_cache,          \
+                     (const CT*)v_cache, block_tables, seq_lens,              \
+                     (short*)out, part_o, part_ml, kvh, bs, max_blocks,       \
+                     part_tokens, scale)
 #define LAUNCH(QPG)                                                           \
   do {                                                                        \
-    hipLaunchKernelGGL((paged_attention_kernel<QPG, true>), grid, block, 0,   \
-                       stream, (const short*)q, (const short*)k_cache,        \
-                       (const short*)v_cache, block_tables, seq_lens,         \
-                       (short*)out, part_o, part_ml, kvh, bs, max_blocks,     \
-                       part_tokens, scale);                                   \
+    if (kv_fp8) LAUNCH_CT(QPG, unsigned char);                                \
+    else LAUNCH_CT(QPG, short);                                               \
     hipLaunchKernelGGL((paged_attention_combine_kernel<QPG>), cgrid, cblock,  \
                        0, stream, part_o, part_ml, (short*)out, kvh,          \
                        n_parts);                                              \

@@ -37,17 +37,24 @@ def block_hashes(prompt_tokens: Sequence[int], block_size: int = BLOCK_SIZE,
 
 
 class KVPool:
+    """cache_dtype: torch.bfloat16 or torch.float8_e4m3fn (OCP fp8 — halves
+    KV bytes and doubles effective decode-attention bandwidth; the gfx950
+    kernels convert with the hardware v_cvt_*_fp8 ops, hip_common.h)."""
+
     def __init__(self, config: ModelConfig, num_blocks: int,
                  device: torch.device, dtype: torch.dtype = torch.bfloat16,
-                 block_size: int = BLOCK_SIZE):
+                 block_size: int = BLOCK_SIZE,
+                 cache_dtype: torch.dtype = None):
         self.cfg = config
         self.num_blocks = num_blocks
         self.block_size = block_size
         self.device = device
         self.dtype = dtype
+        self.cache_dtype = cache_dtype or dtype
         self.tensor = torch.zeros(
             (config.num_layers, 2, num_blocks, config.num_kv_heads,
-             block_size, config.head_dim), dtype=dtype, device=device)
+             block_size, config.head_dim), dtype=self.cache_dtype,
+            device=device)
 
     def layer(self, li: int):
         return self.tensor[li, 0], self.tensor[li, 1]

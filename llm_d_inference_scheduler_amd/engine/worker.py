@@ -163,6 +163,7 @@ class EngineWorker:
                  max_model_len: int = 8192,
                  ttft_slo_ms: float = None,
                  prefix_caching: bool = True,
+                 kv_cache_dtype: str = "auto",
                  seed: int = 0):
         self.cfg = config
         self.device = torch.device(device)
@@ -172,10 +173,17 @@ class EngineWorker:
         self.max_decode_batch = max_decode_batch
         self.max_model_len = max_model_len
         self.ttft_slo_ms = ttft_slo_ms
+        cache_dtype = {"auto": dtype, "bf16": torch.bfloat16,
+                       "fp8": torch.float8_e4m3fn}[kv_cache_dtype]
+        if cache_dtype == torch.float8_e4m3fn and \
+                self.device.type != "cuda":
+            cache_dtype = dtype     # fp8 path needs the gfx950 kernels
         if kv_blocks is None:
-            kv_blocks = KVPool.blocks_for_budget(config, kv_budget_bytes,
-                                                 dtype_bytes=2)
-        self.pool = KVPool(config, kv_blocks, self.device, dtype)
+            kv_blocks = KVPool.blocks_for_budget(
+                config, kv_budget_bytes,
+                dtype_bytes=cache_dtype.itemsize)
+        self.pool = KVPool(config, kv_blocks, self.device, dtype,
+                           cache_dtype=cache_dtype)
         self.mgr = BlockManager(kv_blocks, self.pool.block_size,
                                 prefix_caching=prefix_caching)
         self.model = LlamaRunner(config, self.device, dtype, seed=seed)

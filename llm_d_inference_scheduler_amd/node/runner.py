@@ -82,6 +82,7 @@ class NodeConfig:
     dtype: Any = torch.bfloat16
     kv_blocks: Optional[int] = None
     kv_budget_bytes: int = 8 << 30
+    kv_cache_dtype: str = "auto"    # auto|bf16|fp8 (fp8 = OCP e4m3 cache)
     prefill_chunk_tokens: int = 8192
     max_decode_batch: int = 256
     flow_control: bool = False
@@ -134,7 +135,8 @@ class NodeRunner:
             kv_blocks=cfg.kv_blocks, kv_budget_bytes=cfg.kv_budget_bytes,
             dtype=cfg.dtype, prefill_chunk_tokens=cfg.prefill_chunk_tokens,
             max_decode_batch=cfg.max_decode_batch,
-            ttft_slo_ms=cfg.ttft_slo_ms, seed=cfg.seed)
+            ttft_slo_ms=cfg.ttft_slo_ms,
+            kv_cache_dtype=cfg.kv_cache_dtype, seed=cfg.seed)
         # encode role: vision tower + URL-deduped embedding cache
         self.encoder = None
         if (self.my_spec.role & Role.ENCODE) and cfg.model.vision_hidden:
@@ -624,7 +626,8 @@ class NodeRunner:
                 if local is None:
                     # keep the P2P matched: receive into scratch and drop
                     scratch = self.transfer._staging(n)
-                    torch.distributed.recv(scratch, src=src,
+                    torch.distributed.recv(self.transfer._wire(scratch),
+                                           src=src,
                                            group=self.cfg.transfer_group)
                     self._outbox.append({"type": "done",
                                          "req_id": job["req_id"],

@@ -42,6 +42,11 @@ class KVTransferEngine:
         return torch.empty((n_blocks, L, 2) + tuple(self.pool.shape[3:]),
                            dtype=self.pool.dtype, device=self.device)
 
+    @staticmethod
+    def _wire(t: torch.Tensor) -> torch.Tensor:
+        """RCCL has no fp8 dtype: ship fp8 staging buffers as uint8."""
+        return t.view(torch.uint8) if t.dtype == torch.float8_e4m3fn else t
+
     def send_blocks(self, dst_rank: int, block_ids: List[int]) -> None:
         t0 = time.monotonic()
         ids = torch.tensor(block_ids, dtype=torch.int32, device=self.device)
@@ -49,11 +54,11 @@ class KVTransferEngine:
         if self.on_gpu:
             with torch.cuda.stream(self.stream):
                 ops.move_blocks(self.pool, staging, ids, is_scatter=False)
-                dist.send(staging, dst=dst_rank, group=self.group)
+                dist.send(self._wire(staging), dst=dst_rank, group=self.group)
             self.stream.synchronize()
         else:
             ops.move_blocks(self.pool, staging, ids, is_scatter=False)
-            dist.send(staging, dst=dst_rank, group=self.group)
+            dist.send(self._wire(staging), dst=dst_rank, group=self.group)
         nbytes = staging.numel() * staging.element_size()
         prom.xgmi_kv_transfer_bytes.labels("send").inc(nbytes)
         prom.xgmi_kv_transfer_seconds.observe(time.monotonic() - t0)
@@ -64,11 +69,11 @@ class KVTransferEngine:
         staging = self._staging(len(block_ids))
         if self.on_gpu:
             with torch.cuda.stream(self.stream):
-                dist.recv(staging, src=src_rank, group=self.group)
+                dist.recv(self._wire(staging), src=src_rank, group=self.group)
                 ops.move_blocks(self.pool, staging, ids, is_scatter=True)
             self.stream.synchronize()
         else:
-            dist.recv(staging, src=src_rank, group=self.group)
+            dist.recv(self._wire(staging), src=src_rank, group=self.group)
             ops.move_blocks(self.pool, staging, ids, is_scatter=True)
         nbytes = staging.numel() * staging.element_size()
         prom.xgmi_kv_transfer_bytes.labels("recv").inc(nbytes)

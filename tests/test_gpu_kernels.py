@@ -326,3 +326,113 @@ class TestGpuPrefixIndex:
             hq = rc.hash_tokens(q, 16, 256, seed0)
             expect = host.match_longest(hq, 4)
             assert list(match[i]) == list(expect), i
+
+
+class TestFp8KVCacheGPU:
+    """fp8 (OCP e4m3) KV-cache kernels vs fp32 reference computed on the
+    SAME quantized values — isolates kernel correctness from quantization."""
+
+    def _setup(self, B, seqs_max, qpg=4):
+        KVH, D, BS = 8, 128, 16
+        QH = KVH * qpg
+        max_blocks = (seqs_max + BS - 1) // BS
+        NB = max_blocks * B + 1
+        torch.manual_seed(11)
+        q = torch.randn(B, QH, D, device="cuda").bfloat16()
+        kc8 = torch.randn(NB, KVH, BS, D, device="cuda").to(
+            torch.float8_e4m3fn)
+        vc8 = torch.randn(NB, KVH, BS, D, device="cuda").to(
+            torch.float8_e4m3fn)
+        bt = torch.arange(1, NB, dtype=torch.int32,
+                          device="cuda").view(B, max_blocks)
+        return q, kc8, vc8, bt, KVH, D, BS
+
+    def test_reshape_and_cache_fp8(self, ext):
+        T, KVH, D, NB, BS = 40, 8, 128, 16, 16
+        k = torch.randn(T, KVH, D, device="cuda").bfloat16()
+        v = torch.randn(T, KVH, D, device="cuda").bfloat16()
+        kc = torch.zeros(NB, KVH, BS, D, device="cuda").to(
+            torch.float8_e4m3fn)
+        vc = torch.zeros_like(kc)
+        slots = torch.randperm(NB * BS)[:T].to(torch.int64).cuda()
+        ext.reshape_and_cache(k, v, kc, vc, slots)
+        # the hardware v_cvt_pk_fp8_f32 pack should round like torch's cast
+        # (both round-to-nearest-even); compare exactly on scattered slots
+        want_k = k.float().to(torch.float8_e4m3fn).float().cpu()
+        want_v = v.float().to(torch.float8_e4m3fn).float().cpu()
+        for t in range(T):
+            s = int(slots[t])
+            blk, row = s // BS, s % BS
+            assert torch.equal(kc[blk, :, row].float().cpu(), want_k[t]), t
+            assert torch.equal(vc[blk, :, row].float().cpu(), want_v[t]), t
+
+    def test_paged_attention_fp8(self, ext):
+        from llm_d_inference_scheduler_amd.ops import ref
+        seqs = [77, 1024, 333]
+        q, kc8, vc8, bt, KVH, D, BS = self._setup(len(seqs), max(seqs))
+        sl = torch.tensor(seqs, dtype=torch.int32)
+        out = ext.paged_attention(q, kc8, vc8, bt, sl.cuda(), D ** -0.5)
+        out_ref = ref.paged_attention(to_f32(q), kc8.float().cpu(),
+                                      vc8.float().cpu(), bt.cpu(), sl,
+                                      D ** -0.5)
+        assert torch.allclose(to_f32(out), out_ref, atol=3e-2, rtol=3e-2), \
+            (to_f32(out) - out_ref).abs().max()
+        # split path agrees
+        out2 = ext.paged_attention_split(q, kc8, vc8, bt, sl.cuda(), 4, 512,
+                                         D ** -0.5)
+        assert torch.allclose(to_f32(out), to_f32(out2), atol=2e-2,
+                              rtol=2e-2)
+
+    def test_flash_prefill_fp8(self, ext):
+        KVH, D, BS, qpg = 8, 128, 16, 4
+        QH = KVH * qpg
+        chunk, prior = 333, 171
+        ctx = chunk + prior
+        nb = (ctx + BS - 1) // BS
+        torch.manual_seed(12)
+        q = torch.randn(chunk, QH, D, device="cuda").bfloat16()
+        kc8 = torch.randn(nb + 1, KVH, BS, D, device="cuda").to(
+            torch.float8_e4m3fn)
+        vc8 = torch.randn(nb + 1, KVH, BS, D, device="cuda").to(
+            torch.float8_e4m3fn)
+        bt = torch.arange(1, nb + 1, dtype=torch.int32,
+                          device="cuda").view(1, -1)
+        meta = torch.tensor([[0, chunk, prior]], dtype=torch.int32,
+                            device="cuda")
+        tiles = torch.tensor([(0, v) for v in range(0, chunk * qpg, 128)],
+                             dtype=torch.int32, device="cuda")
+        scale = D ** -0.5
+        out = ext.flash_prefill(q, kc8, vc8, bt, meta, tiles, scale)
+        kk = kc8.float().cpu()[1:].permute(1, 0, 2, 3).reshape(
+            KVH, -1, D)[:, :ctx]
+        vv = vc8.float().cpu()[1:].permute(1, 0, 2, 3).reshape(
+            KVH, -1, D)[:, :ctx]
+        qi = to_f32(q).view(chunk, KVH, qpg, D).permute(1, 2, 0, 3)
+        s = torch.einsum("hgtd,hsd->hgts", qi, kk) * scale
+        t_idx = torch.arange(chunk).view(1, 1, -1, 1)
+        s_idx = torch.arange(ctx).view(1, 1, 1, -1)
+        s.masked_fill_(s_idx > t_idx + prior, float("-inf"))
+        o = torch.einsum("hgts,hsd->hgtd", torch.softmax(s, -1), vv)
+        out_ref = o.permute(2, 0, 1, 3).reshape(chunk, QH, D)
+        assert torch.allclose(to_f32(out), out_ref, atol=3e-2, rtol=3e-2), \
+            (to_f32(out) - out_ref).abs().max()
+
+    def test_engine_fp8_end_to_end(self):
+        from llm_d_inference_scheduler_amd.engine import (EngineRequest,
+                                                          EngineWorker)
+        from llm_d_inference_scheduler_amd.models.configs import ModelConfig
+        cfg = ModelConfig(name="gpu-test", vocab_size=2048, hidden_size=1024,
+                          intermediate_size=2048, num_layers=2, num_heads=8,
+                          num_kv_heads=2, head_dim=128, rope_theta=10000.0)
+        w = EngineWorker(cfg, "cuda:0", kv_blocks=256, dtype=torch.bfloat16,
+                         kv_cache_dtype="fp8", seed=9)
+        assert w.pool.tensor.dtype == torch.float8_e4m3fn
+        w.add_request(EngineRequest("a", prompt_tokens=list(range(300, 430)),
+                                    max_tokens=6))
+        toks = []
+        for _ in range(30):
+            for o in w.step():
+                toks.extend(o.new_tokens)
+            if not w.has_work:
+                break
+        assert len(toks) == 6

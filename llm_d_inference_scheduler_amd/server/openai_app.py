@@ -148,4 +148,30 @@ def build_app(service: NodeService,
         return JSONResponse(status_code=200 if ready else 503,
                             content={"ready": ready})
 
+    @app.get("/readyz")
+    async def readyz():
+        # readiness = pool synced + engine constructed (health.go:52)
+        ready = service.node.datastore.pool_ready() and \
+            service.node.engine is not None
+        return JSONResponse(status_code=200 if ready else 503,
+                            content={"ready": ready})
+
+    @app.get("/debug/pprof/profile")
+    async def pprof_profile(seconds: float = 1.0):
+        """pprof analog (reference --enable-pprof, runner.go:318-324):
+        cProfile the serving loop for N seconds, return pstats text."""
+        import cProfile
+        import io
+        import pstats
+        import time as _t
+        prof = cProfile.Profile()
+        prof.enable()
+        end = _t.time() + min(seconds, 10.0)
+        while _t.time() < end:
+            await __import__("asyncio").sleep(0.05)
+        prof.disable()
+        buf = io.StringIO()
+        pstats.Stats(prof, stream=buf).sort_stats("cumulative").print_stats(40)
+        return Response(content=buf.getvalue(), media_type="text/plain")
+
     return app

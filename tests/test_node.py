@@ -178,3 +178,54 @@ class TestChunkedDecode:
             assert got[0].usage.completion_tokens == 8
             results[chunk] = got[0].tokens
         assert results[None] == results[3]
+
+
+def _pd4_worker(rank, world_size, init_file, out_file):
+    import torch.distributed as dist
+    dist.init_process_group("gloo", init_method=f"file://{init_file}",
+                            rank=rank, world_size=world_size)
+    try:
+        cfg = NodeConfig(model=TINY_LLAMA, rank=rank, world_size=world_size,
+                         topology="pd:1p3d", device="cpu",
+                         dtype=torch.float32, kv_blocks=256,
+                         epp_yaml=EPP_YAML_TIGHT_DISAGG, seed=3)
+        node = NodeRunner(cfg)
+        results = []
+        if rank == 0:
+            for i in range(9):
+                node.submit(make_req(i, n_prompt=48, max_tokens=4))
+        for _ in range(300):
+            node.step()
+            if rank == 0:
+                results.extend(node.drain_completions())
+                done = torch.tensor([1 if len(results) >= 9 else 0])
+            else:
+                done = torch.tensor([0])
+            dist.broadcast(done, src=0)
+            if done.item():
+                break
+        if rank == 0:
+            with open(out_file, "w") as f:
+                json.dump([{ "id": c.request_id, "error": c.error,
+                             "completion": c.usage.completion_tokens}
+                           for c in results], f)
+        node.shutdown()
+        dist.barrier()
+    finally:
+        dist.destroy_process_group()
+
+
+@pytest.mark.timeout(240)
+class TestMultiProcessPD4:
+    def test_pd_1p3d_world4(self, tmp_path):
+        """4-process 1p3d: requests spread over 3 decode ranks, KV from the
+        shared prefill rank — the shape the driver's 8-GPU run scales up."""
+        init_file = str(tmp_path / "pg4_init")
+        out_file = str(tmp_path / "out4.json")
+        mp.start_processes(_pd4_worker, args=(4, init_file, out_file),
+                           nprocs=4, join=True, start_method="spawn")
+        with open(out_file) as f:
+            results = json.load(f)
+        assert len(results) == 9
+        assert all(not r["error"] for r in results), results
+        assert all(r["completion"] == 4 for r in results)

@@ -46,8 +46,8 @@ def parse_args():
                    help="fraction of prompt shared within a request group")
     p.add_argument("--group", type=int, default=4,
                    help="requests per shared-prefix group")
-    p.add_argument("--kv-gb", type=float, default=48.0,
-                   help="KV pool budget per GPU (GB)")
+    p.add_argument("--kv-gb", type=float, default=160.0,
+                   help="KV pool budget per GPU (GB; MI355X has 288)")
     p.add_argument("--kv-dtype", default="auto",
                    choices=["auto", "bf16", "fp8"],
                    help="KV cache storage dtype (fp8 = OCP e4m3; compute "

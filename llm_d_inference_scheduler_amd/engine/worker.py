@@ -430,10 +430,11 @@ class EngineWorker:
         tokens = self._sample_host(logits, [r.temperature for r in finishing])
         now = time.time()
         for req, tok in zip(finishing, tokens):
-            req.first_token_t = now
-            if self.ttft_slo_ms is not None and req.arrival_t and \
-                    (now - req.arrival_t) * 1e3 > self.ttft_slo_ms:
-                req.slo_ok = False
+            if not req.first_token_t:     # preempted reqs keep their TTFT
+                req.first_token_t = now
+                if self.ttft_slo_ms is not None and req.arrival_t and \
+                        (now - req.arrival_t) * 1e3 > self.ttft_slo_ms:
+                    req.slo_ok = False
             self.waiting.remove(req)
             if req.prefill_only:
                 outputs.append(RequestOutput(
@@ -496,7 +497,14 @@ class EngineWorker:
 
         finals = [self._finalize(r) for r in done_now]
         if not active:
-            return self._collect_pending() + finals
+            outs = self._collect_pending() + finals
+            # KV exhaustion with every decodable sequence stalled: nothing
+            # will ever free blocks — preempt the youngest running request
+            # (vLLM-style recompute; the prefix cache usually resurrects
+            # its just-freed blocks, so the recompute is mostly free)
+            if self.running and self.mgr.free_blocks == 0:
+                self._preempt(self.running[-1])
+            return outs
 
         if upd_rows:
             st.bt[torch.tensor(upd_rows, dtype=torch.int64, device=self.device),
@@ -577,6 +585,24 @@ class EngineWorker:
                 self._by_id.pop(req.request_id, None)
             outputs.append(out)
         return outputs
+
+    def _preempt(self, req: EngineRequest) -> None:
+        """Evict a running request and requeue it for recompute: its
+        generation so far folds into the prompt (prefill of prompt+generated
+        produces the next token's logits, so accounting continues exactly).
+        Pending tokens must be collected first (req.inflight == 0)."""
+        assert req.inflight == 0, "collect pending before preempting"
+        log.info("preempting for KV space", req=req.request_id,
+                 generated=len(req.generated))
+        self.running.remove(req)
+        self.dstate.leave(req.request_id)
+        self.mgr.free(req.request_id)
+        merged = (req.prompt_tokens + req.generated)[:self.max_model_len - 1]
+        req.prompt_tokens = merged
+        req.computed = 0
+        req.block_hashes = None
+        req.registered_blocks = 0
+        self.waiting.insert(0, req)
 
     def _finalize(self, req: EngineRequest) -> RequestOutput:
         """Emit the finished output for a request with no in-flight token

@@ -193,3 +193,43 @@ class TestPipelinedDecode:
         fins = {o.request_id: o.completion_tokens
                 for o in outs if o.finished}
         assert fins == {f"r{i}": 2 + 2 * i for i in range(5)}
+
+
+class TestPreemption:
+    def test_kv_exhaustion_preempts_and_completes(self):
+        """Pool too small for all requests at once: the engine preempts
+        (recompute with generation folded into the prompt) instead of
+        deadlocking, and every request still finishes with exact counts."""
+        w = make_worker(kv_blocks=12)           # 192 tokens of KV
+        for i in range(4):
+            w.add_request(EngineRequest(
+                f"r{i}", list(range(3 + i, 43 + i)), max_tokens=30))
+        outs = run_to_completion(w, max_steps=600)
+        fins = {o.request_id: o for o in outs if o.finished}
+        assert len(fins) == 4, fins.keys()
+        for o in fins.values():
+            assert o.completion_tokens == 30
+        assert w.mgr.usage == pytest.approx(0.0)
+
+    def test_preempted_greedy_tokens_unchanged(self):
+        """A preempted+recomputed request produces the same greedy tokens
+        as an undisturbed run."""
+        prompt = list(range(11, 51))
+        w = make_worker(kv_blocks=256)
+        w.add_request(EngineRequest("ref", list(prompt), max_tokens=12))
+        ref = [o for o in run_to_completion(w) if o.finished][0].all_tokens
+
+        w2 = make_worker(kv_blocks=256)
+        w2.add_request(EngineRequest("p", list(prompt), max_tokens=12))
+        for _ in range(6):
+            w2.step()
+        # force-preempt mid-generation
+        w2.step()
+        outs = list(w2._collect_pending())
+        w2._preempt(w2.running[-1])
+        for _ in range(200):
+            outs.extend(w2.step())
+            if not w2.has_work:
+                break
+        got = [o for o in outs if o.finished][0].all_tokens
+        assert got == ref

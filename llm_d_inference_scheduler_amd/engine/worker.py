@@ -164,12 +164,16 @@ class EngineWorker:
                  ttft_slo_ms: float = None,
                  prefix_caching: bool = True,
                  kv_cache_dtype: str = "auto",
+                 prefill_min_tokens: int = 4096,
+                 prefill_max_delay_ms: float = 60.0,
                  seed: int = 0):
         self.cfg = config
         self.device = torch.device(device)
         self.role = role
         self.dtype = dtype
         self.prefill_chunk_tokens = prefill_chunk_tokens
+        self.prefill_min_tokens = prefill_min_tokens
+        self.prefill_max_delay_ms = prefill_max_delay_ms
         self.max_decode_batch = max_decode_batch
         self.max_model_len = max_model_len
         self.ttft_slo_ms = ttft_slo_ms
@@ -271,7 +275,7 @@ class EngineWorker:
         one decode pass over the running batch. Decode token values surface
         one step late (pipelined readback)."""
         outputs: List[RequestOutput] = []
-        if self.waiting:
+        if self.waiting and self._should_prefill():
             outputs.extend(self._prefill_pass())
         if self.running:
             outputs.extend(self._decode_pass())
@@ -279,6 +283,22 @@ class EngineWorker:
             outputs.extend(self._collect_pending())
         self.steps += 1
         return outputs
+
+    def _should_prefill(self) -> bool:
+        """Batch prefill work: a small per-step chunk (one trickling
+        arrival) launches a half-empty grid every layer; accumulating
+        arrivals to `prefill_min_tokens` fills the chip and amortizes the
+        pass, bounded by `prefill_max_delay_ms` of added TTFT."""
+        if not self.running:
+            return True                    # decode idle (or prefill role)
+        if self.waiting[0].computed > 0:
+            return True                    # finish a split chunk promptly
+        pending = sum(r.prompt_len - r.computed for r in self.waiting)
+        if pending >= self.prefill_min_tokens:
+            return True
+        oldest = self.waiting[0].arrival_t
+        return oldest > 0 and \
+            (time.time() - oldest) * 1e3 >= self.prefill_max_delay_ms
 
     # ---- prefill ----
     def _admit_prompt(self, req: EngineRequest) -> None:

@@ -320,7 +320,11 @@ class NodeRunner:
         """One hash_prompts + one match_longest launch for the whole
         admission batch (SURVEY.md §2.6 MI355X mapping). Returns per-request
         precomputed-attribute dicts, or None to use the host path."""
-        if (self._gpu_prefix is None or len(batch) < 2 or
+        # small admission batches route through the C++ host index: the GPU
+        # batch path's D2H sync would stall routing behind the whole queued
+        # engine step (steady state has 1-2 arrivals/step; the two-launch
+        # GPU path pays off on admission bursts)
+        if (self._gpu_prefix is None or len(batch) < 8 or
                 any(not r.prompt_tokens for r in batch)):
             return None
         from ..datalayer.attributes import (PREFIX_CACHE_MATCH_INFO as KEY,

@@ -37,8 +37,10 @@ def parse_args():
                    help="mono=prefix-aware DP decode; pd=P/D disagg; "
                         "fc=flow control at overload; epd=E/P/D multimodal")
     p.add_argument("--model", default="llama-3-8b")
+    # the reference's regression harness shape: input 1024 / output 1024
+    # (config/manifests/regression-testing/single-workload-regression.yaml)
     p.add_argument("--prompt-len", type=int, default=1024)
-    p.add_argument("--max-tokens", type=int, default=128)
+    p.add_argument("--max-tokens", type=int, default=1024)
     p.add_argument("--concurrency", type=int, default=128,
                    help="in-flight requests per decode rank (closed loop; "
                         "128 is the measured single-GPU throughput knee)")

@@ -128,3 +128,45 @@ class HeaderBasedTestingFilter(Filter):
         names = set(want.split(","))
         return [ep for ep in endpoints
                 if ep.name in names or ep.metadata.address in names]
+
+
+@register_plugin("utilization-detector")
+class UtilizationDetectorPlugin(Filter):
+    """The utilization saturation detector registered as a plugin: usable
+    from EndpointPickerConfig both as the flow-control saturation signal
+    and as a scheduling Filter with fail-open fallback
+    (saturationdetector/utilization/README.md)."""
+
+    def __init__(self, name: str = "", **params):
+        super().__init__(name, **params)
+        from ..flowcontrol.saturation import UtilizationSaturationDetector
+        self.detector = UtilizationSaturationDetector(
+            queue_threshold=float(params.get("queueDepthThreshold", 5.0)),
+            kv_threshold=float(params.get("kvCacheUtilThreshold", 0.8)),
+            staleness_s=float(params.get("metricsStalenessSeconds", 0.5)))
+
+    def filter(self, ctx, endpoints):
+        return self.detector.filter(ctx, endpoints)
+
+    def is_saturated(self, endpoints) -> bool:
+        return self.detector.is_saturated(endpoints)
+
+
+@register_plugin("concurrency-detector")
+class ConcurrencyDetectorPlugin(Filter):
+    """Aggregate in-flight-fraction saturation detector
+    (saturationdetector/concurrency); passes endpoints through unchanged
+    when used in a profile."""
+
+    def __init__(self, name: str = "", **params):
+        super().__init__(name, **params)
+        from ..flowcontrol.saturation import ConcurrencySaturationDetector
+        self.detector = ConcurrencySaturationDetector(
+            max_inflight_per_endpoint=int(params.get(
+                "maxInflightPerEndpoint", 256)))
+
+    def filter(self, ctx, endpoints):
+        return endpoints
+
+    def is_saturated(self, endpoints) -> bool:
+        return self.detector.is_saturated(endpoints)

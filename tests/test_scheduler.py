@@ -167,3 +167,35 @@ class TestSchedulerLoop:
         sched = self._disagg_scheduler(non_cached_tokens=512)
         res = sched.schedule(ctx, endpoints)
         assert "prefill" not in res.profile_results
+
+
+class TestSaturationDetectorPlugins:
+    def test_utilization_filter_fail_open(self):
+        from llm_d_inference_scheduler_amd.datalayer.datastore import \
+            make_endpoint
+        from llm_d_inference_scheduler_amd.datalayer.endpoint import Metrics
+        from llm_d_inference_scheduler_amd.plugins.registry import global_registry
+        f = global_registry.instantiate("utilization-detector",
+                                 queueDepthThreshold=5,
+                                 metricsStalenessSeconds=1e9)
+        eps = [make_endpoint(f"gpu{i}", i) for i in range(3)]
+        eps[0].update_metrics(Metrics(waiting_queue_size=0))
+        eps[1].update_metrics(Metrics(waiting_queue_size=10))   # saturated
+        eps[2].update_metrics(Metrics(waiting_queue_size=2))
+        kept = f.filter(None, eps)
+        assert [e.name for e in kept] == ["gpu0", "gpu2"]
+        # all saturated -> fail-open returns everything
+        for ep in eps:
+            ep.update_metrics(Metrics(waiting_queue_size=50))
+        assert len(f.filter(None, eps)) == 3
+        assert f.is_saturated(eps)
+
+    def test_concurrency_detector_plugin(self):
+        from llm_d_inference_scheduler_amd.datalayer.datastore import \
+            make_endpoint
+        from llm_d_inference_scheduler_amd.plugins.registry import global_registry
+        d = global_registry.instantiate("concurrency-detector",
+                                 maxInflightPerEndpoint=1)
+        eps = [make_endpoint("gpu0", 0)]
+        assert d.filter(None, eps) == eps
+        assert not d.is_saturated(eps)

@@ -30,7 +30,12 @@ def build_app(service: NodeService,
               tokenizer: Optional[HashTokenizer] = None) -> FastAPI:
     app = FastAPI(title="llm-d-inference-scheduler-amd")
     mux = ParserMux()
+    from ..handlers.parsers import VertexAIParser, VllmGrpcParser
+    mux.register("application/grpc", VllmGrpcParser())
+    mux.register("application/grpc+proto", VllmGrpcParser())
+    mux.register("application/vnd.vertex-ai+json", VertexAIParser())
     tok = tokenizer or HashTokenizer()
+    app.state.service = service
 
     def _error(status: int, reason: str, detail: str = "") -> JSONResponse:
         return JSONResponse(
@@ -118,6 +123,13 @@ def build_app(service: NodeService,
     @app.post("/v1/completions")
     async def completions(request: Request):
         return await _handle(request, "/v1/completions")
+
+    @app.post("/vllm.VllmEngine/{method}")
+    async def vllm_grpc(request: Request, method: str):
+        """vLLM gRPC wire-format front door (parsers/vllmgrpc): bodies are
+        gRPC-framed protobuf; dispatch runs through the same mux/parser
+        plugins keyed by the :path the reference's ext-proc sees."""
+        return await _handle(request, f"/vllm.VllmEngine/{method}")
 
     @app.post("/v1/chat/completions")
     async def chat_completions(request: Request):

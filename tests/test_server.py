@@ -89,3 +89,76 @@ class TestOpenAIServer:
     def test_models(self, client):
         r = client.get("/v1/models")
         assert r.json()["data"][0]["id"] == "tiny-llama"
+
+
+class TestHermeticExtras:
+    def test_model_rewrite_round_trip(self, client):
+        """Weighted InferenceModelRewrite: request under the alias routes to
+        the real model; the response carries the CLIENT-facing name
+        (rewriteModelName back, server.go:471)."""
+        from llm_d_inference_scheduler_amd.api.modelrewrite import (
+            InferenceModelRewrite, RewriteRule, RewriteTarget)
+        node = client.app.state.service.node
+        node.datastore.put_model_rewrite(InferenceModelRewrite(
+            name="alias", rules=[RewriteRule(
+                model="tiny-alias",
+                targets=[RewriteTarget("tiny-llama", weight=1)])]))
+        r = client.post("/v1/completions", json={
+            "model": "tiny-alias", "prompt": "rewrite me please now",
+            "max_tokens": 3})
+        assert r.status_code == 200, r.text
+        assert r.json()["model"] == "tiny-alias"
+
+    def test_vllm_grpc_content_type(self, client):
+        """vLLM gRPC wire bodies route through the mux by content-type."""
+        def ev(v):
+            out = b""
+            while True:
+                b7 = v & 0x7F
+                v >>= 7
+                if v:
+                    out += bytes([b7 | 0x80])
+                else:
+                    return out + bytes([b7])
+        msg = bytes([0x0A]) + ev(10) + b"tiny-llama"
+        packed = b"".join(ev(t) for t in [9, 8, 7, 6, 5, 4, 3, 2, 1])
+        msg += bytes([0x1A]) + ev(len(packed)) + packed
+        msg += bytes([0x20]) + ev(3)
+        body = bytes([0]) + len(msg).to_bytes(4, "big") + msg
+        r = client.post("/vllm.VllmEngine/Generate", content=body,
+                        headers={"content-type": "application/grpc"})
+        assert r.status_code == 200, r.text
+        assert r.json()["usage"]["completion_tokens"] == 3
+
+
+class TestFlowControl429:
+    def test_saturated_returns_429_with_reason(self):
+        cfg = NodeConfig(model=TINY_LLAMA, world_size=1, topology="mono",
+                         device="cpu", dtype=torch.float32, kv_blocks=256,
+                         flow_control=True, fc_global_max_items=2)
+        node = NodeRunner(cfg)
+        service = NodeService(node)
+        service.start()
+        app = build_app(service)
+        # force saturation: the flow queue fills to fc_global_max_items
+        # and further arrivals are rejected with 429 + reason header
+        node.detector.is_saturated = lambda eps: True
+        with TestClient(app) as c:
+            import concurrent.futures as cf
+
+            def post(i):
+                return c.post("/v1/completions", json={
+                    "model": "tiny-llama",
+                    "prompt": "x " * 40 + str(i), "max_tokens": 4})
+            with cf.ThreadPoolExecutor(max_workers=8) as pool:
+                futs = [pool.submit(post, i) for i in range(8)]
+                import time as _t
+                _t.sleep(1.0)
+                node.detector.is_saturated = lambda eps: False  # drain
+                rs = [f.result(timeout=60) for f in futs]
+            codes = sorted(r.status_code for r in rs)
+            rejected = [r for r in rs if r.status_code == 429]
+            assert rejected, codes
+            assert rejected[0].headers.get("x-request-dropped-reason")
+            assert 200 in codes, codes
+        service.stop()

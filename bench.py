@@ -41,9 +41,10 @@ def parse_args():
     # (config/manifests/regression-testing/single-workload-regression.yaml)
     p.add_argument("--prompt-len", type=int, default=1024)
     p.add_argument("--max-tokens", type=int, default=1024)
-    p.add_argument("--concurrency", type=int, default=128,
+    p.add_argument("--concurrency", type=int, default=256,
                    help="in-flight requests per decode rank (closed loop; "
-                        "128 is the measured single-GPU throughput knee)")
+                        "256 is the measured single-GPU knee at the "
+                        "in=1024/out=1024 reference shape)")
     p.add_argument("--shared-prefix", type=float, default=0.5,
                    help="fraction of prompt shared within a request group")
     p.add_argument("--group", type=int, default=4,

@@ -41,9 +41,16 @@ __device__ __forceinline__ int k_swz(int row, int byte_off) {
   return row * 256 + (byte_off ^ ((row & 15) << 4));
 }
 
-// V^T tile LDS: [128 dims][VSTRIDE shorts]; 160-B row stride keeps 16-B
-// alignment and caps ds_read_b128 conflicts at 2-way.
-constexpr int VSTRIDE = 80;  // shorts per V^T row (64 data + 16 pad)
+// V tile is stored ROW-MAJOR with the same XOR swizzle as K; the PV
+// A-operand (V^T fragments) is gathered by the gfx950 hardware transpose
+// read v_ds_read_b64_tr_b16: within each fixed 16-lane group, lane i's
+// element j comes from the group's combined 64-short window at
+// [i + 16*j] — pointing lane i at &V[kv0 + i/4][d0 + (i%4)*4] delivers
+// exactly A[dim=d0+i][kv=kv0+j] (layout verified empirically,
+// tools/tr16_probe.hip). This removes the per-element b16
+// scatter-transpose staging that dominated SQ_LDS_BANK_CONFLICT
+// (profiles/r01_pmc_counters.md).
+typedef __attribute__((ext_vector_type(4))) short short4v;
 
 __device__ __forceinline__ unsigned int pack_bf16(float a, float b) {
   return ((unsigned int)(unsigned short)f32_to_bf16(b) << 16) |
@@ -71,7 +78,7 @@ __global__ __launch_bounds__(NW * WAVE) void flash_prefill_kernel(
   const int32_t* bt = block_tables + (int64_t)seq * max_blocks;
 
   __shared__ short k_lds[KVBLK * D];          // swizzled, byte-addressed
-  __shared__ short v_lds[D * VSTRIDE];        // V^T, padded stride
+  __shared__ short v_lds[KVBLK * D];          // row-major, same swizzle
 
   const int tid = threadIdx.x;
   const int wave = tid / WAVE;
@@ -125,9 +132,7 @@ __global__ __launch_bounds__(NW * WAVE) void flash_prefill_kernel(
         vpiece = load_kv8_bf16(v_cache + base + c16 * 8);
       }
       *(short8*)((char*)k_lds + k_swz(row, c16 * 16)) = piece;
-#pragma unroll
-      for (int j = 0; j < 8; ++j)                     // transpose V
-        v_lds[(c16 * 8 + j) * VSTRIDE + row] = vpiece[j];
+      *(short8*)((char*)v_lds + k_swz(row, c16 * 16)) = vpiece;
     }
     __syncthreads();
 
@@ -193,17 +198,29 @@ __global__ __launch_bounds__(NW * WAVE) void flash_prefill_kernel(
       const bf16x8 p_frag0 = __builtin_bit_cast(bf16x8, f0);
       const bf16x8 p_frag1 = __builtin_bit_cast(bf16x8, f1);
 
-      // ---- O^T += V^T . P^T ----
+      // ---- O^T += V^T . P^T (A-operand via hardware transpose reads) ----
+      const int i16 = lane % 16;
+      const int d_grp = 16 * ((lane & 31) >> 4);   // group's dim base
+      const int dim_col = d_grp + (i16 % 4) * 4;
 #pragma unroll
       for (int dt = 0; dt < D / 32; ++dt) {
-        const bf16x8 v0 = *(const bf16x8*)(
-            &v_lds[(dt * 32 + qcol) * VSTRIDE + hi * 8]);
-        acc_o[dt] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(v0, p_frag0,
-                                                            acc_o[dt], 0, 0, 0);
-        const bf16x8 v1 = *(const bf16x8*)(
-            &v_lds[(dt * 32 + qcol) * VSTRIDE + 16 + hi * 8]);
-        acc_o[dt] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(v1, p_frag1,
-                                                            acc_o[dt], 0, 0, 0);
+#pragma unroll
+        for (int ks = 0; ks < 2; ++ks) {
+          const int kv0 = ks * 16 + hi * 8 + i16 / 4;
+          const int dc = (dt * 32 + dim_col) * 2;
+          short4v lo = __builtin_amdgcn_ds_read_tr16_b64_v4i16(
+              (__attribute__((address_space(3))) short4v*)(
+                  (char*)v_lds + kv0 * 256 + (dc ^ ((kv0 & 15) << 4))));
+          const int kv1 = kv0 + 4;
+          short4v hi4 = __builtin_amdgcn_ds_read_tr16_b64_v4i16(
+              (__attribute__((address_space(3))) short4v*)(
+                  (char*)v_lds + kv1 * 256 + (dc ^ ((kv1 & 15) << 4))));
+          short8 vfrag8 = {lo[0], lo[1], lo[2], lo[3],
+                           hi4[0], hi4[1], hi4[2], hi4[3]};
+          acc_o[dt] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
+              __builtin_bit_cast(bf16x8, vfrag8),
+              ks == 0 ? p_frag0 : p_frag1, acc_o[dt], 0, 0, 0);
+        }
       }
     }
   }

@@ -170,3 +170,33 @@ class ConcurrencyDetectorPlugin(Filter):
 
     def is_saturated(self, endpoints) -> bool:
         return self.detector.is_saturated(endpoints)
+
+
+@register_plugin("destination-endpoint-served-verifier")
+class DestinationEndpointServedVerifier(Filter):
+    """Conformance plugin (reference
+    .../requestcontrol/test/responsereceived, registered in the production
+    runner for IGW conformance, runner.go:496-499): verifies that the
+    endpoint which SERVED the response (`x-gateway-destination-endpoint-
+    served` response header) matches the scheduler's pick; counts
+    mismatches for the conformance harness. As a Filter it passes
+    endpoints through unchanged."""
+
+    SERVED_HEADER = "x-gateway-destination-endpoint-served"
+
+    def __init__(self, name: str = "", **params):
+        super().__init__(name, **params)
+        self.checked = 0
+        self.mismatches = 0
+
+    def filter(self, ctx, endpoints):
+        return endpoints
+
+    def response_received(self, ctx, target, headers) -> None:
+        served = (headers or {}).get(self.SERVED_HEADER, "")
+        if not served:
+            return
+        self.checked += 1
+        if target is not None and served not in (target.name,
+                                                 target.metadata.address):
+            self.mismatches += 1

@@ -140,3 +140,17 @@ class TestDirector:
             d, Usage(prompt_tokens=40, completion_tokens=16, ttft_ms=30,
                      tpot_ms=9))
         assert load.snapshot()[0] == 0
+
+
+class TestServedVerifier:
+    def test_served_header_verification(self):
+        from llm_d_inference_scheduler_amd.plugins.registry import \
+            global_registry
+        from llm_d_inference_scheduler_amd.datalayer.datastore import \
+            make_endpoint
+        v = global_registry.instantiate("destination-endpoint-served-verifier")
+        ep = make_endpoint("gpu1", 1)
+        v.response_received(None, ep, {v.SERVED_HEADER: "gpu1"})
+        v.response_received(None, ep, {v.SERVED_HEADER: "gpu0"})
+        v.response_received(None, ep, {})
+        assert (v.checked, v.mismatches) == (2, 1)

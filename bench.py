@@ -259,13 +259,12 @@ def main():
     def drain():
         if rank != 0:
             return
+        stats["ttft_ms"].extend(node.drain_ttft_events())
         for c in node.drain_completions():
             if c.error:
                 stats["errors"] += 1
                 continue
             stats["completed"] += 1
-            if c.usage.ttft_ms is not None:
-                stats["ttft_ms"].append(c.usage.ttft_ms)
             if c.usage.e2e_ms is not None:
                 stats["e2e_ms"].append(c.usage.e2e_ms)
 
@@ -285,7 +284,7 @@ def main():
         node.step()
         drain()
     if rank == 0:
-        node.epp_latencies.clear()  # only count timed-region decisions
+        stats["ttft_ms"].clear()    # TTFT stats start at the timed region
 
     # ---- timed region: EXACTLY K steps ----
     sync()

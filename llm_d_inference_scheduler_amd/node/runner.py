@@ -159,6 +159,7 @@ class NodeRunner:
         # decode-side: req_ids whose tokens stream back per step
         self._streaming_ids: set = set()
         self._token_events: List[tuple] = []  # rank0: (req_id, [tokens])
+        self._ttft_events: List[float] = []   # rank0: ttft_ms at first token
         self._step = 0
 
         self.is_router = (self.rank == 0)
@@ -246,6 +247,12 @@ class NodeRunner:
         assert self.is_router
         out = self._completions
         self._completions = []
+        return out
+
+    def drain_ttft_events(self) -> List[float]:
+        assert self.is_router
+        out = self._ttft_events
+        self._ttft_events = []
         return out
 
     def drain_token_events(self) -> List[tuple]:
@@ -445,6 +452,8 @@ class NodeRunner:
                 if self._precise is not None:
                     self._precise.apply_events(f"gpu{m['src']}", m["s"],
                                                m["e"])
+            elif t == "ttft" and self.is_router:
+                self._ttft_events.append(m["ms"])
             elif t == "tokens" and self.is_router:
                 self._token_events.append((m["req_id"], m["toks"]))
                 decision = self._decisions.get(m["req_id"])
@@ -662,6 +671,16 @@ class NodeRunner:
             elif out.kind == "embedding" and out.finished:
                 self._emit_done(out, tokens=[])
             else:
+                if out.ttft_ms is not None and not out.finished:
+                    # first decode token: report TTFT now, not only at
+                    # completion (long generations would otherwise starve
+                    # the bench's TTFT stats window)
+                    msg = {"type": "ttft", "req_id": out.request_id,
+                           "ms": out.ttft_ms}
+                    if self.is_router:
+                        self._ttft_events.append(out.ttft_ms)
+                    else:
+                        self._outbox.append(msg)
                 if out.request_id in self._streaming_ids and out.new_tokens:
                     msg = {"type": "tokens", "req_id": out.request_id,
                            "toks": list(out.new_tokens)}

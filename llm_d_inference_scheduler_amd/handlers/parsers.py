@@ -8,8 +8,9 @@ requesthandling/types.go:65).
   parse on the response side (parsers/openai/openai.go:34-139).
 * passthrough-parser: raw bytes, Skip -> fallback random-endpoint routing.
 * vertexai-parser: Google Vertex AI payload shape.
-* vllm-grpc-parser: N/A on this node — the reference proxies an external
-  vLLM gRPC engine; here the engines are in-process (docs/PARITY.md).
+* vllm-grpc-parser: direct protobuf wire-format decode of the VllmEngine
+  service's Generate/Embed requests + usage responses (no codegen);
+  non-routable methods Skip to random-endpoint fallback.
 """
 import json
 import uuid
@@ -171,6 +172,8 @@ class VertexAIParser(Parser):
             data = json.loads(body)
         except (ValueError, TypeError) as e:
             return ParseResult(error=f"invalid JSON: {e}")
+        if not isinstance(data, dict):
+            return ParseResult(error="body is not a JSON object")
         model = str(data.get("model", "")) or headers.get("x-vertex-model", "")
         if not model:
             return ParseResult(error="missing model")
@@ -180,8 +183,11 @@ class VertexAIParser(Parser):
         contents = data.get("contents") or []
         parts: List[str] = []
         for c in contents if isinstance(contents, list) else []:
-            for part in c.get("parts", []):
-                if "text" in part:
+            if not isinstance(c, dict):
+                continue
+            parts_list = c.get("parts", [])
+            for part in parts_list if isinstance(parts_list, list) else []:
+                if isinstance(part, dict) and "text" in part:
                     parts.append(str(part["text"]))
         req.prompt = "\n".join(parts)
         gen_cfg = data.get("generationConfig") or {}

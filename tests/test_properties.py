@@ -1,0 +1,147 @@
+"""Property-based invariants (hypothesis): BlockManager accounting under
+random op sequences, flow-queue ordering, parser robustness on arbitrary
+bytes. ADVICE-tier hardening: these are the data structures whose
+accounting bugs would corrupt KV or strand requests silently."""
+import string
+
+from hypothesis import HealthCheck, given, settings
+from hypothesis import strategies as st
+
+from llm_d_inference_scheduler_amd.engine.kvcache import (BlockManager,
+                                                          block_hashes)
+
+SMALL = settings(max_examples=60, deadline=None,
+                 suppress_health_check=[HealthCheck.too_slow])
+
+
+@st.composite
+def op_sequences(draw):
+    n_ops = draw(st.integers(10, 60))
+    ops = []
+    for i in range(n_ops):
+        kind = draw(st.sampled_from(["prompt", "grow", "free", "take"]))
+        ops.append((kind, draw(st.integers(0, 7)),
+                    draw(st.integers(1, 70))))
+    return ops
+
+
+class TestBlockManagerInvariants:
+    @SMALL
+    @given(op_sequences(), st.integers(8, 64))
+    def test_accounting_never_breaks(self, ops, num_blocks):
+        mgr = BlockManager(num_blocks)
+        live = {}          # seq -> token count requested
+        taken = []
+        prompts = {s: list(range(10 + s, 10 + s + 64)) for s in range(8)}
+        for kind, s, n in ops:
+            sid = f"s{s}"
+            if kind == "prompt" and sid not in mgr.tables:
+                h = block_hashes(prompts[s][:n])
+                matched = mgr.allocate_prompt(sid, h, max(1, n))
+                assert matched <= max(0, n - 1)
+                if not mgr.allocate(sid, n):
+                    mgr.free(sid)
+                    continue
+                live[sid] = n
+                full = n // 16
+                for b in range(full):
+                    mgr.register_block(sid, b, int(h[b]))
+            elif kind == "grow" and sid in mgr.tables:
+                cur = live.get(sid, 0)
+                if mgr.allocate(sid, cur + n):
+                    live[sid] = cur + n
+            elif kind == "free" and sid in mgr.tables:
+                mgr.free(sid)
+                live.pop(sid, None)
+            elif kind == "take":
+                got = mgr.take_blocks(min(n, 4))
+                if got is not None:
+                    taken.extend(got)
+                    if len(taken) > 8:
+                        mgr.release_blocks(taken)
+                        taken = []
+            # --- invariants after every op ---
+            allocated = set()
+            for t in mgr.tables.values():
+                for blk in t:
+                    allocated.add(blk)
+            # a block is never in two tables unless refcounted > 1
+            refsum = sum(mgr._refcnt)
+            table_refs = sum(len(t) for t in mgr.tables.values())
+            assert refsum == table_refs + len(taken)
+            assert 0 <= mgr.free_blocks <= num_blocks
+            for blk in allocated:
+                assert mgr._refcnt[blk] >= 1
+                assert blk not in mgr._free_lru
+        for sid in list(mgr.tables):
+            mgr.free(sid)
+        mgr.release_blocks(taken)
+        assert mgr.free_blocks == num_blocks
+
+    @SMALL
+    @given(st.lists(st.integers(0, 2 ** 31 - 1), min_size=0, max_size=100))
+    def test_block_hashes_deterministic_prefix_property(self, tokens):
+        h1 = block_hashes(tokens)
+        h2 = block_hashes(tokens)
+        assert list(h1) == list(h2)
+        if len(tokens) >= 32:
+            # chained property: a longer prompt with identical prefix
+            # shares exactly the leading block hashes
+            h3 = block_hashes(tokens + [1, 2, 3, 4])
+            assert list(h3[:len(h1)]) == list(h1)
+
+
+class TestQueueProperties:
+    @SMALL
+    @given(st.lists(st.tuples(st.integers(0, 1 << 30), st.integers(1, 999)),
+                    min_size=1, max_size=80))
+    def test_maxminheap_orders_by_key(self, items):
+        from llm_d_inference_scheduler_amd import _router_core as rc
+        h = rc.MaxMinHeap()
+        for i, (key, _) in enumerate(items):
+            h.push(i, float(key))
+        keys = []
+        for _ in range(len(items)):
+            popped = h.pop()        # (id, key, bytes) of the min-key item
+            assert popped is not None
+            keys.append(float(items[int(popped[0])][0]))
+        assert keys == sorted(keys)
+        assert h.pop() is None
+
+    @SMALL
+    @given(st.lists(st.integers(0, 1 << 20), min_size=1, max_size=60))
+    def test_listqueue_fifo(self, ids):
+        from llm_d_inference_scheduler_amd import _router_core as rc
+        q = rc.ListQueue()
+        for pos, i in enumerate(ids):
+            q.push(pos, 0.0, i)
+        out = [int(q.pop()[0]) for _ in range(len(ids))]
+        assert out == list(range(len(ids)))     # strict FIFO by insertion
+        assert q.pop() is None
+
+
+class TestParserFuzz:
+    @SMALL
+    @given(st.binary(min_size=0, max_size=300))
+    def test_openai_parser_never_crashes(self, body):
+        from llm_d_inference_scheduler_amd.handlers.parsers import \
+            OpenAIParser
+        res = OpenAIParser().parse_request(body, {}, "/v1/completions")
+        assert res.error is not None or res.request is not None or res.skip
+
+    @SMALL
+    @given(st.binary(min_size=0, max_size=300))
+    def test_vllm_grpc_parser_never_crashes(self, body):
+        from llm_d_inference_scheduler_amd.handlers.parsers import \
+            VllmGrpcParser
+        res = VllmGrpcParser().parse_request(
+            body, {}, "/vllm.VllmEngine/Generate")
+        assert res.error is not None or res.request is not None or res.skip
+
+    @SMALL
+    @given(st.text(alphabet=string.printable, max_size=200))
+    def test_vertexai_parser_never_crashes(self, text):
+        from llm_d_inference_scheduler_amd.handlers.parsers import \
+            VertexAIParser
+        res = VertexAIParser().parse_request(text.encode(), {}, "/v1")
+        assert res.error is not None or res.request is not None or res.skip

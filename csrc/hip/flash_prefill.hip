@@ -26,7 +26,7 @@
 namespace {
 
 constexpr int D = 128;       // head_dim
-constexpr int KVBLK = 32;    // tokens per KV tile
+constexpr int KVBLK = 64;    // tokens per KV tile (2 mfma sub-tiles)
 constexpr int NW = 4;        // waves per workgroup
 constexpr int QROWS = 32;    // q virtual-rows per wave
 constexpr float NEG = -1e30f;
@@ -58,7 +58,7 @@ __device__ __forceinline__ unsigned int pack_bf16(float a, float b) {
 }
 
 template <int QPG, typename CT>
-__global__ __launch_bounds__(NW * WAVE) void flash_prefill_kernel(
+__global__ __launch_bounds__(NW * WAVE, 2) void flash_prefill_kernel(
     const short* __restrict__ q,        // [T, QH, D]
     const CT* __restrict__ k_cache,     // [NB, KVH, BS, D] bf16|fp8
     const CT* __restrict__ v_cache,     // [NB, KVH, BS, D] bf16|fp8
@@ -79,6 +79,7 @@ __global__ __launch_bounds__(NW * WAVE) void flash_prefill_kernel(
 
   __shared__ short k_lds[KVBLK * D];          // swizzled, byte-addressed
   __shared__ short v_lds[KVBLK * D];          // row-major, same swizzle
+  static_assert(KVBLK * 16 % (NW * WAVE) == 0, "staging divides evenly");
 
   const int tid = threadIdx.x;
   const int wave = tid / WAVE;
@@ -116,11 +117,11 @@ __global__ __launch_bounds__(NW * WAVE) void flash_prefill_kernel(
   const int kv_end_w = prior + p_max_w + 1;
 
   for (int tile0 = 0; tile0 < kv_end_wg; tile0 += KVBLK) {
-    // ---- stage K (swizzled) and V^T into LDS: 512 16-B pieces ----
+    // ---- stage K (swizzled) and V into LDS ----
     __syncthreads();  // previous tile's reads complete before overwrite
 #pragma unroll
-    for (int rep = 0; rep < 2; ++rep) {
-      const int ch = tid + rep * (NW * WAVE);        // 0..511
+    for (int rep = 0; rep < KVBLK * 16 / (NW * WAVE); ++rep) {
+      const int ch = tid + rep * (NW * WAVE);
       const int row = ch / 16, c16 = ch % 16;
       const int t = tile0 + row;
       short8 piece = {0, 0, 0, 0, 0, 0, 0, 0};
@@ -138,22 +139,27 @@ __global__ __launch_bounds__(NW * WAVE) void flash_prefill_kernel(
 
     if (tile0 >= kv_end_w) continue;  // beyond this wave's causal horizon
 
+#pragma unroll 1
+    for (int half = 0; half < 2; ++half) {
+    const int sub0 = tile0 + 32 * half;
+    if (sub0 >= kv_end_w) break;
     // ---- S^T = K . Q^T : acc rows = kv tokens, cols = q rows ----
     f32x16 acc_s = (f32x16)(0.f);
 #pragma unroll
     for (int kk = 0; kk < D / 16; ++kk) {
       const bf16x8 k_frag = *(const bf16x8*)(
-          (const char*)k_lds + k_swz(qcol, (kk * 16 + hi * 8) * 2));
+          (const char*)k_lds + k_swz(32 * half + qcol,
+                                     (kk * 16 + hi * 8) * 2));
       acc_s = __builtin_amdgcn_mfma_f32_32x32x16_bf16(k_frag, q_frag[kk],
                                                       acc_s, 0, 0, 0);
     }
 
-    // ---- per-lane online softmax over this tile's 32 kv tokens ----
+    // ---- per-lane online softmax over this sub-tile's 32 kv tokens ----
     float s[16];
     float tmax = NEG;
 #pragma unroll
     for (int r = 0; r < 16; ++r) {
-      const int kv = tile0 + (r & 3) + 8 * (r >> 2) + 4 * hi;
+      const int kv = sub0 + (r & 3) + 8 * (r >> 2) + 4 * hi;
       const bool valid = active && kv <= prior + p;
       s[r] = valid ? acc_s[r] * scale : NEG;
       tmax = fmaxf(tmax, s[r]);
@@ -206,7 +212,7 @@ __global__ __launch_bounds__(NW * WAVE) void flash_prefill_kernel(
       for (int dt = 0; dt < D / 32; ++dt) {
 #pragma unroll
         for (int ks = 0; ks < 2; ++ks) {
-          const int kv0 = ks * 16 + hi * 8 + i16 / 4;
+          const int kv0 = 32 * half + ks * 16 + hi * 8 + i16 / 4;
           const int dc = (dt * 32 + dim_col) * 2;
           short4v lo = __builtin_amdgcn_ds_read_tr16_b64_v4i16(
               (__attribute__((address_space(3))) short4v*)(
@@ -223,6 +229,7 @@ __global__ __launch_bounds__(NW * WAVE) void flash_prefill_kernel(
         }
       }
     }
+    }  // half
   }
 
   // ---- epilogue: out[seq_start+p][qh][dim] = acc_o / l ----

@@ -34,12 +34,22 @@ class Span:
 
 
 class Tracer:
+    """Env config mirrors the reference's OTEL_* driven init
+    (pkg/telemetry/tracing.go): LLMD_TRACING=0 disables span recording;
+    LLMD_TRACE_EXPORT=<path> appends finished spans as JSONL at shutdown
+    (the OTLP-exporter stand-in — there is no network egress here)."""
+
     def __init__(self, service_name: str, capacity: int = 4096):
+        import atexit
+        import os
         self.service_name = service_name
         self._spans: deque = deque(maxlen=capacity)
         self._lock = threading.Lock()
         self._local = threading.local()
-        self.enabled = True
+        self.enabled = os.environ.get("LLMD_TRACING", "1") != "0"
+        export = os.environ.get("LLMD_TRACE_EXPORT", "")
+        if export and self.enabled:
+            atexit.register(self.export_jsonl, export)
 
     @contextmanager
     def span(self, name: str, **attrs):

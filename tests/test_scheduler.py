@@ -199,3 +199,26 @@ class TestSaturationDetectorPlugins:
         eps = [make_endpoint("gpu0", 0)]
         assert d.filter(None, eps) == eps
         assert not d.is_saturated(eps)
+
+
+class TestTracerExport:
+    def test_env_export_jsonl(self, tmp_path, monkeypatch):
+        import importlib
+        monkeypatch.setenv("LLMD_TRACE_EXPORT", str(tmp_path / "spans.jsonl"))
+        from llm_d_inference_scheduler_amd.telemetry import tracing
+        t = tracing.Tracer("test-svc")
+        with t.span("gateway.request", model="m") as s:
+            s.set_attribute("k", 1)
+        t.export_jsonl(str(tmp_path / "spans.jsonl"))
+        import json
+        lines = (tmp_path / "spans.jsonl").read_text().strip().splitlines()
+        rec = json.loads(lines[-1])
+        assert rec["name"] == "gateway.request"
+
+    def test_tracing_disabled_env(self, monkeypatch):
+        monkeypatch.setenv("LLMD_TRACING", "0")
+        from llm_d_inference_scheduler_amd.telemetry.tracing import Tracer
+        t = Tracer("svc")
+        with t.span("x") as s:
+            s.set_attribute("a", 1)   # noop span accepts attributes
+        assert not t.finished_spans()

@@ -243,6 +243,17 @@ def main():
 
     stats = {"completed": 0, "errors": 0, "epp_ms": [], "ttft_ms": [],
              "e2e_ms": []}
+    # fc mode: per-priority-tier accounting (BASELINE config 4 is about
+    # holding the critical tier's SLO while sheddables are dropped)
+    tiers = ["critical", "standard", "batch"]
+    tier_stats = {t: {"completed": 0, "shed": 0, "ttft_ms": []}
+                  for t in tiers}
+
+    def tier_of(req_id):
+        try:
+            return tiers[int(req_id.split("-")[-1]) % 3]
+        except (ValueError, IndexError):
+            return "standard"
 
     ramp = max(1, target_inflight // max(8, args.warmup // 2))
 
@@ -263,8 +274,15 @@ def main():
         for c in node.drain_completions():
             if c.error:
                 stats["errors"] += 1
+                if fc:
+                    tier_stats[tier_of(c.request_id)]["shed"] += 1
                 continue
             stats["completed"] += 1
+            if fc:
+                ts = tier_stats[tier_of(c.request_id)]
+                ts["completed"] += 1
+                if c.usage.ttft_ms is not None:
+                    ts["ttft_ms"].append(c.usage.ttft_ms)
             if c.usage.e2e_ms is not None:
                 stats["e2e_ms"].append(c.usage.e2e_ms)
 
@@ -347,6 +365,10 @@ def main():
                 "p99_ttft_ms": p99_ttft,
                 "total_tok_s": round(tokens / elapsed, 2),
                 "errors": stats["errors"],
+                **({"tiers": {t: {
+                    "completed": v["completed"], "shed": v["shed"],
+                    "p50_ttft_ms": _pctl(v["ttft_ms"], 50)}
+                    for t, v in tier_stats.items()}} if fc else {}),
             },
         }
         print(json.dumps(result), flush=True)

@@ -357,7 +357,8 @@ def main():
                 "mode": args.mode,
                 "prompt_len": args.prompt_len,
                 "max_tokens": args.max_tokens,
-                "kv_cache_dtype": args.kv_dtype,
+                "kv_cache_dtype": str(node.engine.pool.cache_dtype
+                                      ).replace("torch.", ""),
                 "ttft_slo_ms": args.ttft_slo_ms,
                 "routed_req_s": round(completed / elapsed, 2),
                 "p50_epp_latency_ms": p50_epp,

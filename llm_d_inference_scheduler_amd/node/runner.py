@@ -249,6 +249,34 @@ class NodeRunner:
         self._completions = []
         return out
 
+    def cancel(self, request_id: str) -> None:
+        """Stream-death cleanup (server.go:246-253 deferred forced
+        response-complete): abort a routed request everywhere — pending
+        arrivals, decisions, engines on every rank, transfer state."""
+        assert self.is_router
+        self._arrivals = [r for r in self._arrivals
+                          if r.request_id != request_id]
+        self._chunked.pop(request_id, None)
+        decision = self._decisions.pop(request_id, None)
+        self._outbox.append({"type": "abort", "req_id": request_id})
+        self._abort_local(request_id)
+        if decision is not None:
+            # unwind response hooks (inflight counters, training state)
+            self.director.handle_response_complete(decision, Usage())
+
+    def _abort_local(self, rid: str) -> None:
+        self.engine.abort(rid)
+        self._deferred_prefill.pop(rid, None)
+        self._handoff_dst.pop(rid, None)
+        self._streaming_ids.discard(rid)
+        pend = self._pending_adoption.pop(rid, None)
+        if pend and pend.get("reserved"):
+            self.engine.mgr.release_blocks(pend["reserved"])
+        self._emb_pending.pop(rid, None)
+        self._awaiting_embeds.pop(rid, None)
+        self._encode_jobs = [j for j in self._encode_jobs
+                             if j["req_id"] != rid]
+
     def drain_ttft_events(self) -> List[float]:
         assert self.is_router
         out = self._ttft_events
@@ -441,6 +469,8 @@ class NodeRunner:
                 self._handle_assign(m)
             elif t == "done" and self.is_router:
                 self._handle_done(m)
+            elif t == "abort" and not self.is_router:
+                self._abort_local(m["req_id"])
             elif t == "need_prefill" and m.get("dst") == self.rank:
                 req = self._deferred_prefill.pop(m["req_id"], None)
                 if req is not None:

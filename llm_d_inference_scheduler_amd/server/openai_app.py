@@ -63,6 +63,7 @@ def build_app(service: NodeService,
                                      media_type="text/event-stream")
         completion = await asyncio.to_thread(handle.wait, 120.0)
         if completion is None:
+            service.cancel(req.request_id)     # reclaim engine resources
             return _error(504, "timeout")
         if completion.error:
             status = 429 if completion.error in (
@@ -99,16 +100,23 @@ def build_app(service: NodeService,
 
     async def _sse_stream(req, handle, t0):
         idx = 0
-        while True:
-            item = await asyncio.to_thread(handle.token_queue.get)
-            if item is None:
-                break
-            chunk = {"id": f"cmpl-{req.request_id}", "object":
-                     "text_completion.chunk", "model": req.model,
-                     "choices": [{"index": 0, "text": tok.decode([item]),
-                                  "finish_reason": None}]}
-            idx += 1
-            yield f"data: {json.dumps(chunk)}\n\n"
+        try:
+            while True:
+                item = await asyncio.to_thread(handle.token_queue.get)
+                if item is None:
+                    break
+                chunk = {"id": f"cmpl-{req.request_id}", "object":
+                         "text_completion.chunk", "model": req.model,
+                         "choices": [{"index": 0,
+                                      "text": tok.decode([item]),
+                                      "finish_reason": None}]}
+                idx += 1
+                yield f"data: {json.dumps(chunk)}\n\n"
+        except asyncio.CancelledError:
+            # client disconnected mid-stream: stream-death cleanup
+            # (server.go:246-253) — abort across the node
+            service.cancel(req.request_id)
+            raise
         completion = handle.completion
         if completion is not None and not completion.error:
             final = {"id": f"cmpl-{req.request_id}",

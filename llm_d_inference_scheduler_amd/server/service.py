@@ -43,6 +43,7 @@ class NodeService:
         self.node = node
         self.step_interval_s = step_interval_s
         self._handles: Dict[str, RequestHandle] = {}
+        self._cancels: "queue.Queue" = queue.Queue()
         self._lock = threading.Lock()
         self._stop = threading.Event()
         self._thread: Optional[threading.Thread] = None
@@ -54,7 +55,24 @@ class NodeService:
         self.node.submit(req)
         return handle
 
+    def cancel(self, request_id: str) -> None:
+        """Client went away (stream death): finish the handle immediately
+        and unwind the request across the node on the loop thread."""
+        with self._lock:
+            h = self._handles.pop(request_id, None)
+        if h is not None:
+            from ..handlers.parsers import Usage
+            h.finish(Completion(request_id=request_id, usage=Usage(),
+                                error="canceled"))
+        self._cancels.put(request_id)
+
     def step_once(self) -> None:
+        while True:
+            try:
+                rid = self._cancels.get_nowait()
+            except queue.Empty:
+                break
+            self.node.cancel(rid)
         self.node.step()
         for req_id, toks in self.node.drain_token_events():
             with self._lock:

@@ -162,3 +162,29 @@ class TestFlowControl429:
             assert rejected[0].headers.get("x-request-dropped-reason")
             assert 200 in codes, codes
         service.stop()
+
+
+class TestCancellation:
+    def test_cancel_unwinds_engine(self):
+        """Stream-death cleanup: canceling a routed request frees engine
+        state everywhere (server.go:246-253 forced-complete analog)."""
+        cfg = NodeConfig(model=TINY_LLAMA, world_size=1, topology="mono",
+                         device="cpu", dtype=torch.float32, kv_blocks=256)
+        node = NodeRunner(cfg)
+        service = NodeService(node)
+        from llm_d_inference_scheduler_amd.scheduling.types import LLMRequest
+        req = LLMRequest(request_id="long", model="tiny-llama", prompt="",
+                         prompt_tokens=list(range(5, 69)), max_tokens=5000)
+        h = service.submit(req)
+        for _ in range(6):
+            service.step_once()
+        assert node.engine.has_work
+        service.cancel("long")
+        for _ in range(4):
+            service.step_once()
+        done = h.wait(1.0)
+        assert done is not None and done.error == "canceled"
+        assert not node.engine.has_work
+        assert node.inflight == 0
+        assert node.engine.mgr.usage == pytest.approx(0.0)
+        node.shutdown()

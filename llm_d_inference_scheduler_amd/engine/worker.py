@@ -39,6 +39,7 @@ class EngineRequest:
     is_embedding: bool = False
     prefill_only: bool = False      # disagg: stop after prefill + 1st token
     cached_tokens: int = 0          # prefix-cache reuse (engine-measured)
+    priority: int = 0               # InferenceObjective priority (preemption)
     # multimodal: embeddings for the first len(prefix_embeds) prompt rows
     # (prompt_tokens must carry placeholder ids for those positions)
     prefix_embeds: "Optional[torch.Tensor]" = None
@@ -535,7 +536,10 @@ class EngineWorker:
             # (vLLM-style recompute; the prefix cache usually resurrects
             # its just-freed blocks, so the recompute is mostly free)
             if self.running and self.mgr.free_blocks == 0:
-                self._preempt(self.running[-1])
+                # victim: lowest InferenceObjective priority, youngest last
+                victim = min(reversed(self.running),
+                             key=lambda r: r.priority)
+                self._preempt(victim)
             return outs
 
         if upd_rows:

@@ -414,6 +414,7 @@ class NodeRunner:
             msg = {"type": "assign", "req_id": req.request_id,
                    "dst": decode_rank,
                    "tokens": req.prompt_tokens or [],
+                   "priority": req.priority,
                    "max_tokens": max_tokens,
                    "temperature": req.temperature,
                    "is_embedding": req.is_embedding,
@@ -499,6 +500,7 @@ class NodeRunner:
             max_tokens=m["max_tokens"], temperature=m["temperature"],
             is_embedding=m.get("is_embedding", False),
             cached_tokens=m.get("cached", 0),
+            priority=m.get("priority", 0),
             arrival_t=m.get("arrival") or 0.0)
         # the rank that runs the prompt (prefill stage or monolithic decode)
         prompt_rank = prefill_rank if (
@@ -526,6 +528,7 @@ class NodeRunner:
                 request_id=m["req_id"], prompt_tokens=list(m["tokens"]),
                 max_tokens=m["max_tokens"], temperature=m["temperature"],
                 cached_tokens=m.get("cached", 0),
+                priority=m.get("priority", 0),
                 arrival_t=m.get("arrival") or 0.0)
             local_hit = False
             if shared_storage:

@@ -233,3 +233,23 @@ class TestPreemption:
                 break
         got = [o for o in outs if o.finished][0].all_tokens
         assert got == ref
+
+    def test_preemption_prefers_low_priority(self):
+        """Victim selection honors InferenceObjective priority: the
+        sheddable request is recomputed, the critical one is untouched."""
+        w = make_worker(kv_blocks=8)
+        w.add_request(EngineRequest("crit", list(range(5, 37)),
+                                    max_tokens=40, priority=10))
+        w.add_request(EngineRequest("shed", list(range(200, 232)),
+                                    max_tokens=40, priority=-1))
+        preempted = []
+        orig = w._preempt
+
+        def spy(req):
+            preempted.append(req.request_id)
+            return orig(req)
+        w._preempt = spy
+        outs = run_to_completion(w, max_steps=800)
+        fins = {o.request_id for o in outs if o.finished}
+        assert fins == {"crit", "shed"}
+        assert preempted and all(r == "shed" for r in preempted)

@@ -217,6 +217,23 @@ class NodeRunner:
                 self._approx = p
             elif isinstance(p, PrecisePrefixCacheScorer):
                 self._precise = p
+
+        # endpoint lifecycle events fan out to plugin remove hooks — the
+        # notification-source surface that drives the reference's precise
+        # prefix ZMQ subscriber add/remove (datalayer/source/notifications,
+        # precise_prefix_cache.go:622-691)
+        def _on_endpoint_event(kind: str, ep) -> None:
+            if kind != "remove":
+                return
+            for plugin in self.loaded.plugins.values():
+                hook = getattr(plugin, "remove_endpoint", None)
+                if hook is not None:
+                    try:
+                        hook(ep)
+                    except Exception as e:  # pragma: no cover
+                        log.error("remove_endpoint hook failed",
+                                  plugin=plugin.name, err=str(e))
+        self.datastore.on_endpoint_event(_on_endpoint_event)
         if self._approx is not None and str(cfg.device).startswith("cuda"):
             from ..ops.prefix import GpuPrefixIndex
             self._gpu_prefix = GpuPrefixIndex(cfg.device)

@@ -135,3 +135,32 @@ schedulingProfiles:
             [int(x) for x in h], ["gpu0"])
         assert m["gpu0"] >= 3
         node.shutdown()
+
+
+class TestEndpointLifecycleNotifications:
+    def test_removal_fans_out_to_plugin_hooks(self):
+        """Endpoint removal drives plugin remove_endpoint hooks — the
+        notification-source surface that the reference uses to tear down
+        per-pod ZMQ subscribers (precise_prefix_cache.go:622-691)."""
+        from llm_d_inference_scheduler_amd.node import NodeConfig, NodeRunner
+        yaml_cfg = """
+plugins:
+  - type: decode-filter
+  - type: precise-prefix-cache-scorer
+  - type: queue-scorer
+  - type: max-score-picker
+schedulingProfiles:
+  - name: decode
+    plugins:
+      - {pluginRef: decode-filter}
+      - {pluginRef: precise-prefix-cache-scorer, weight: 2}
+      - {pluginRef: max-score-picker}
+"""
+        node = NodeRunner(NodeConfig(model=TINY_LLAMA, epp_yaml=yaml_cfg,
+                                     device="cpu", dtype=torch.float32,
+                                     kv_blocks=64))
+        node._precise.apply_events("gpu0", [111, 222], [])
+        assert node._precise.index.size == 2
+        node.datastore.remove_endpoint("gpu0")
+        assert node._precise.index.size == 0
+        node.shutdown()

@@ -473,3 +473,55 @@ class TestPagedAttentionV3:
                                     D ** -0.5)
         assert torch.allclose(to_f32(base), to_f32(v3), atol=2e-2,
                               rtol=2e-2), (to_f32(base) - to_f32(v3)).abs().max()
+
+
+class TestEngineFeaturesGPU:
+    def _cfg(self):
+        from llm_d_inference_scheduler_amd.models.configs import ModelConfig
+        return ModelConfig(name="gpu-test", vocab_size=2048,
+                           hidden_size=1024, intermediate_size=2048,
+                           num_layers=2, num_heads=8, num_kv_heads=2,
+                           head_dim=128, rope_theta=10000.0)
+
+    def test_chunked_decode_node_gpu(self):
+        from llm_d_inference_scheduler_amd.node import NodeConfig, NodeRunner
+        from llm_d_inference_scheduler_amd.scheduling.types import LLMRequest
+        results = {}
+        for chunk in (None, 4):
+            node = NodeRunner(NodeConfig(
+                model=self._cfg(), world_size=1, topology="mono",
+                device="cuda:0", dtype=torch.bfloat16, kv_blocks=512,
+                decode_chunk_tokens=chunk, seed=7))
+            node.submit(LLMRequest(request_id="r", model="gpu-test",
+                                   prompt="", prompt_tokens=list(range(300,
+                                                                       400)),
+                                   max_tokens=10))
+            got = []
+            for _ in range(200):
+                node.step()
+                got.extend(node.drain_completions())
+                if got:
+                    break
+            node.shutdown()
+            assert got and not got[0].error
+            assert got[0].usage.completion_tokens == 10
+            results[chunk] = got[0].tokens
+        assert results[None] == results[4]
+
+    def test_preemption_gpu(self):
+        from llm_d_inference_scheduler_amd.engine import (EngineRequest,
+                                                          EngineWorker)
+        w = EngineWorker(self._cfg(), "cuda:0", kv_blocks=24,
+                         dtype=torch.bfloat16, seed=5)
+        for i in range(3):
+            w.add_request(EngineRequest(f"r{i}",
+                                        list(range(100 + i, 180 + i)),
+                                        max_tokens=40))
+        outs = []
+        for _ in range(600):
+            outs.extend(w.step())
+            if not w.has_work:
+                break
+        fins = {o.request_id: o for o in outs if o.finished}
+        assert len(fins) == 3
+        assert all(o.completion_tokens == 40 for o in fins.values())

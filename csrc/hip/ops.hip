@@ -48,6 +48,10 @@ hipError_t lds_flash_prefill(const void*, const void*, const void*,
                              const int32_t*, const int32_t*, const int32_t*,
                              void*, int, int, int, int, int, int, int, float,
                              hipStream_t);
+hipError_t lds_flash_prefill_glds(const void*, const void*, const void*,
+                                  const int32_t*, const int32_t*,
+                                  const int32_t*, void*, int, int, int, int,
+                                  int, int, float, hipStream_t);
 }
 
 namespace {
@@ -255,6 +259,15 @@ torch::Tensor flash_prefill(torch::Tensor q, torch::Tensor k_cache,
   int qh = (int)q.size(1), d = (int)q.size(2);
   int kvh = (int)k_cache.size(1), bs = (int)k_cache.size(2);
   auto out = torch::empty_like(q);
+  // bf16 KV rides the glds staging pipeline; fp8 needs convert-on-stage
+  if (!kv_fp8_flag(k_cache) && block_tables.size(1) <= 1024) {
+    CHECK_HIP(lds_flash_prefill_glds(
+        q.data_ptr(), k_cache.data_ptr(), v_cache.data_ptr(),
+        block_tables.data_ptr<int32_t>(), seq_meta.data_ptr<int32_t>(),
+        tiles.data_ptr<int32_t>(), out.data_ptr(), (int)tiles.size(0), qh,
+        kvh, bs, d, (int)block_tables.size(1), (float)scale, cur_stream()));
+    return out;
+  }
   CHECK_HIP(lds_flash_prefill(
       q.data_ptr(), k_cache.data_ptr(), v_cache.data_ptr(),
       block_tables.data_ptr<int32_t>(), seq_meta.data_ptr<int32_t>(),

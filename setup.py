@@ -36,6 +36,7 @@ hip_ops = cpp_extension.CUDAExtension(
         "csrc/hip/kv_cache.hip",
         "csrc/hip/paged_attention.hip",
         "csrc/hip/flash_prefill.hip",
+        "csrc/hip/flash_prefill_glds.hip",
     ],
     extra_compile_args={
         "cxx": ["-O3"],

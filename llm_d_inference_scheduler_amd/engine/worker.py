@@ -40,6 +40,7 @@ class EngineRequest:
     prefill_only: bool = False      # disagg: stop after prefill + 1st token
     cached_tokens: int = 0          # prefix-cache reuse (engine-measured)
     priority: int = 0               # InferenceObjective priority (preemption)
+    stop_token_ids: Optional[List[int]] = None  # finish_reason "stop"
     # multimodal: embeddings for the first len(prefix_embeds) prompt rows
     # (prompt_tokens must carry placeholder ids for those positions)
     prefix_embeds: "Optional[torch.Tensor]" = None
@@ -73,6 +74,7 @@ class RequestOutput:
     request_id: str
     new_tokens: List[int]
     finished: bool = False
+    finish_reason: str = "length"   # length | stop
     kind: str = "decode"            # decode | prefill_done | embedding
     # prefill_done payload (disagg hand-off):
     kv_blocks: Optional[List[int]] = None
@@ -602,9 +604,13 @@ class EngineWorker:
             self.total_generated += 1
             if req.slo_ok:
                 self.total_generated_slo += 1
-            finished = len(req.generated) >= req.max_tokens
+            stopped = bool(req.stop_token_ids) and \
+                int(tok) in req.stop_token_ids
+            finished = stopped or len(req.generated) >= req.max_tokens
             out = RequestOutput(request_id=req.request_id,
-                                new_tokens=[int(tok)], finished=finished)
+                                new_tokens=[int(tok)], finished=finished,
+                                finish_reason="stop" if stopped
+                                else "length")
             if finished:
                 n_gen = len(req.generated)
                 out.all_tokens = list(req.generated)

@@ -119,6 +119,8 @@ class OpenAIParser(Parser):
         req.max_tokens = int(data.get("max_tokens",
                                       data.get("max_completion_tokens", 16)))
         req.temperature = float(data.get("temperature", 0.0))
+        if isinstance(data.get("stop_token_ids"), list):
+            req.stop_token_ids = [int(t) for t in data["stop_token_ids"]]
         req.streaming = bool(data.get("stream", False))
         req.session_id = str(data.get("user", "") or "")
         _headers_into(req, headers)

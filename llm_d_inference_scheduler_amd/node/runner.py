@@ -119,6 +119,7 @@ class Completion:
     request_id: str
     usage: Usage
     tokens: List[int] = field(default_factory=list)
+    finish_reason: str = "length"   # length | stop
     error: str = ""
 
 
@@ -432,6 +433,7 @@ class NodeRunner:
                    "dst": decode_rank,
                    "tokens": req.prompt_tokens or [],
                    "priority": req.priority,
+                   "stop": req.stop_token_ids,
                    "max_tokens": max_tokens,
                    "temperature": req.temperature,
                    "is_embedding": req.is_embedding,
@@ -518,6 +520,7 @@ class NodeRunner:
             is_embedding=m.get("is_embedding", False),
             cached_tokens=m.get("cached", 0),
             priority=m.get("priority", 0),
+            stop_token_ids=m.get("stop"),
             arrival_t=m.get("arrival") or 0.0)
         # the rank that runs the prompt (prefill stage or monolithic decode)
         prompt_rank = prefill_rank if (
@@ -546,6 +549,7 @@ class NodeRunner:
                 max_tokens=m["max_tokens"], temperature=m["temperature"],
                 cached_tokens=m.get("cached", 0),
                 priority=m.get("priority", 0),
+                stop_token_ids=m.get("stop"),
                 arrival_t=m.get("arrival") or 0.0)
             local_hit = False
             if shared_storage:
@@ -626,7 +630,9 @@ class NodeRunner:
         self._chunked.pop(m["req_id"], None)
         self._completions.append(Completion(
             request_id=m["req_id"], usage=usage,
-            tokens=m.get("tokens", []), error=m.get("error", "")))
+            tokens=m.get("tokens", []),
+            finish_reason=m.get("finish_reason", "length"),
+            error=m.get("error", "")))
 
     # ---- encode stage (E/PD, E/P/D) ----
     def _run_encoder(self, max_jobs: int = 8) -> None:
@@ -745,6 +751,7 @@ class NodeRunner:
 
     def _emit_done(self, out: RequestOutput, tokens) -> None:
         msg = {"type": "done", "req_id": out.request_id,
+               "finish_reason": out.finish_reason,
                "prompt_tokens": out.prompt_tokens,
                "completion_tokens": out.completion_tokens,
                "cached_tokens": out.cached_tokens,

@@ -30,6 +30,7 @@ class LLMRequest:
     prompt_tokens: Optional[List[int]] = None      # set by token-producer
     max_tokens: int = 16
     temperature: float = 0.0
+    stop_token_ids: Optional[List[int]] = None
     streaming: bool = False
     is_embedding: bool = False
     mm_items: List[MultiModalItem] = field(default_factory=list)

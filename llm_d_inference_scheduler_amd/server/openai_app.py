@@ -56,6 +56,17 @@ def build_app(service: NodeService,
                           "without an upstream")
         if req.prompt_tokens is None and not req.prompt and req.messages:
             pass  # token-producer will tokenize messages
+        raw_stop = (result.request.raw_body or {}).get("stop") \
+            if result.request.raw_body else None
+        if raw_stop and req.stop_token_ids is None:
+            stops = [raw_stop] if isinstance(raw_stop, str) else raw_stop
+            ids = []
+            for sword in stops[:4]:
+                t = tok(str(sword))
+                if t:
+                    ids.append(t[0])   # single-token stop approximation
+            if ids:
+                req.stop_token_ids = ids
         t0 = time.time()
         handle = service.submit(req)
         if req.streaming and not req.is_embedding:
@@ -84,12 +95,12 @@ def build_app(service: NodeService,
                     "model": req.model,
                     "choices": [{"index": 0, "message":
                                  {"role": "assistant", "content": text},
-                                 "finish_reason": "length"}],
+                                 "finish_reason": completion.finish_reason}],
                     "usage": usage}
         return {"id": f"cmpl-{req.request_id}", "object": "text_completion",
                 "created": created, "model": req.model,
                 "choices": [{"index": 0, "text": text,
-                             "finish_reason": "length"}],
+                             "finish_reason": completion.finish_reason}],
                 "usage": usage}
 
     def _embedding_response(req, completion):

@@ -253,3 +253,22 @@ class TestPreemption:
         fins = {o.request_id for o in outs if o.finished}
         assert fins == {"crit", "shed"}
         assert preempted and all(r == "shed" for r in preempted)
+
+
+class TestStopTokens:
+    def test_stop_token_finishes_early(self):
+        w = make_worker()
+        # run once to learn the greedy continuation, then stop on its
+        # 3rd generated token
+        w.add_request(EngineRequest("probe", list(range(5, 45)),
+                                    max_tokens=8))
+        ref = [o for o in run_to_completion(w) if o.finished][0].all_tokens
+        w2 = make_worker()
+        w2.add_request(EngineRequest("s", list(range(5, 45)), max_tokens=8,
+                                     stop_token_ids=[ref[2]]))
+        outs = run_to_completion(w2)
+        fin = [o for o in outs if o.finished][0]
+        assert fin.finish_reason == "stop"
+        assert fin.completion_tokens == 3
+        assert fin.all_tokens == ref[:3]
+        assert not w2.has_work and w2.mgr.usage == pytest.approx(0.0)

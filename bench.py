@@ -55,6 +55,11 @@ def parse_args():
                    choices=["auto", "bf16", "fp8"],
                    help="KV cache storage dtype (fp8 = OCP e4m3; compute "
                         "stays bf16)")
+    p.add_argument("--arrival-rate", type=float, default=0.0,
+                   help="open-loop Poisson arrivals (req/s, whole node); "
+                        "0 = closed loop at --concurrency. The reference's "
+                        "regression harness sweeps this rate "
+                        "(single-workload-regression.yaml:30-45)")
     p.add_argument("--ttft-slo-ms", type=float, default=2000.0)
     p.add_argument("--device", default=None, help="override (cpu for tests)")
     p.add_argument("--seed", type=int, default=1234)
@@ -256,9 +261,20 @@ def main():
             return "standard"
 
     ramp = max(1, target_inflight // max(8, args.warmup // 2))
+    arrival_state = {"next": None}
 
     def feed(limit=None):
         if rank != 0:
+            return
+        if args.arrival_rate > 0:
+            # open-loop: Poisson arrivals against the wall clock
+            now = time.perf_counter()
+            if arrival_state["next"] is None:
+                arrival_state["next"] = now
+            while arrival_state["next"] <= now:
+                node.submit(workload.next_request())
+                arrival_state["next"] += \
+                    workload.rng.expovariate(args.arrival_rate)
             return
         n = 0
         while node.inflight + len(node._arrivals) < target_inflight:
@@ -355,6 +371,7 @@ def main():
                 "seq_len": args.prompt_len + args.max_tokens,
                 "parallelism": parallelism,
                 "mode": args.mode,
+                "arrival_rate_req_s": args.arrival_rate or None,
                 "prompt_len": args.prompt_len,
                 "max_tokens": args.max_tokens,
                 "kv_cache_dtype": str(node.engine.pool.cache_dtype

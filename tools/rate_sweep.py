@@ -1,0 +1,46 @@
+"""Latency-throughput curve over a request-rate sweep (the reference's
+regression-harness methodology, single-workload-regression.yaml:30-45;
+BASELINE.md "harness shape to reuse"). Run on a GPU box:
+
+  python tools/rate_sweep.py --rates 4 8 12 16 20 --steps 300
+"""
+import argparse
+import json
+import os
+import subprocess
+import sys
+
+ROOT = os.path.join(os.path.dirname(os.path.abspath(__file__)), "..")
+
+
+def main():
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--rates", type=float, nargs="+",
+                    default=[4, 8, 12, 16, 20])
+    ap.add_argument("--steps", type=int, default=300)
+    ap.add_argument("--warmup", type=int, default=60)
+    ap.add_argument("--extra", nargs="*", default=[])
+    args = ap.parse_args()
+    rows = []
+    for r in args.rates:
+        cmd = [sys.executable, os.path.join(ROOT, "bench.py"),
+               "--steps", str(args.steps), "--warmup", str(args.warmup),
+               "--arrival-rate", str(r)] + args.extra
+        out = subprocess.run(cmd, capture_output=True, text=True)
+        line = out.stdout.strip().splitlines()[-1]
+        d = json.loads(line)
+        rows.append((r, d))
+        c = d["config"]
+        print(f"rate {r:6.1f} req/s -> goodput {d['value']:8.1f} tok/s  "
+              f"routed {c['routed_req_s']:6.2f}/s  "
+              f"p50_ttft {c['p50_ttft_ms']}  p99_ttft {c['p99_ttft_ms']}",
+              flush=True)
+    print(json.dumps([{ "rate": r, "goodput": d["value"],
+                        "p50_ttft_ms": d["config"]["p50_ttft_ms"],
+                        "p99_ttft_ms": d["config"]["p99_ttft_ms"],
+                        "routed_req_s": d["config"]["routed_req_s"]}
+                      for r, d in rows]))
+
+
+if __name__ == "__main__":
+    main()

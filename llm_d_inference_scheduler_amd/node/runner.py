@@ -603,7 +603,8 @@ class NodeRunner:
             if state["first_usage"] is None:
                 state["first_usage"] = usage
             orig = state["orig"]
-            if len(state["tokens"]) < orig.max_tokens:
+            if len(state["tokens"]) < orig.max_tokens and \
+                    m.get("finish_reason", "length") != "stop":
                 cont = LLMRequest(
                     request_id=orig.request_id, model=orig.model,
                     prompt="", target_model=orig.target_model,
@@ -611,6 +612,7 @@ class NodeRunner:
                     state["tokens"],
                     max_tokens=orig.max_tokens,
                     temperature=orig.temperature,
+                    stop_token_ids=orig.stop_token_ids,
                     streaming=orig.streaming, headers=dict(orig.headers),
                     objective_name=orig.objective_name,
                     fairness_id=orig.fairness_id)
@@ -625,7 +627,8 @@ class NodeRunner:
                           e2e_ms=(first.e2e_ms or 0) + (usage.e2e_ms or 0))
             self._completions.append(Completion(
                 request_id=m["req_id"], usage=usage,
-                tokens=state["tokens"], error=""))
+                tokens=state["tokens"],
+                finish_reason=m.get("finish_reason", "length"), error=""))
             return
         self._chunked.pop(m["req_id"], None)
         self._completions.append(Completion(

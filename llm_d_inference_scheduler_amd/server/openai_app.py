@@ -134,7 +134,8 @@ def build_app(service: NodeService,
                      "object": "text_completion.chunk",
                      "model": req.model,
                      "choices": [{"index": 0, "text": "",
-                                  "finish_reason": "length"}],
+                                  "finish_reason":
+                                      completion.finish_reason}],
                      "usage": completion.usage.to_openai()}
             yield f"data: {json.dumps(final)}\n\n"
         yield "data: [DONE]\n\n"

@@ -229,3 +229,38 @@ class TestMultiProcessPD4:
         assert len(results) == 9
         assert all(not r["error"] for r in results), results
         assert all(r["completion"] == 4 for r in results)
+
+
+class TestChunkedStop:
+    def test_stop_token_ends_chunked_decode(self):
+        from llm_d_inference_scheduler_amd.scheduling.types import LLMRequest
+        cfg = NodeConfig(model=TINY_LLAMA, world_size=1, topology="mono",
+                         device="cpu", dtype=torch.float32, kv_blocks=256,
+                         seed=5)
+        node = NodeRunner(cfg)
+        node.submit(make_req(0, n_prompt=40, max_tokens=10))
+        ref = None
+        for _ in range(120):
+            node.step()
+            done = node.drain_completions()
+            if done:
+                ref = done[0].tokens
+                break
+        node.shutdown()
+        cfg2 = NodeConfig(model=TINY_LLAMA, world_size=1, topology="mono",
+                          device="cpu", dtype=torch.float32, kv_blocks=256,
+                          decode_chunk_tokens=3, seed=5)
+        node = NodeRunner(cfg2)
+        req = make_req(0, n_prompt=40, max_tokens=10)
+        req.stop_token_ids = [ref[4]]       # stops inside the 2nd chunk
+        node.submit(req)
+        got = None
+        for _ in range(200):
+            node.step()
+            done = node.drain_completions()
+            if done:
+                got = done[0]
+                break
+        node.shutdown()
+        assert got is not None and got.finish_reason == "stop"
+        assert got.tokens == ref[:5]

@@ -188,3 +188,45 @@ class TestCancellation:
         assert node.inflight == 0
         assert node.engine.mgr.usage == pytest.approx(0.0)
         node.shutdown()
+
+
+class TestMetricsGolden:
+    def test_metrics_golden_families(self, client):
+        """Golden family inventory (reference pkg/epp/metrics/testdata/):
+        every documented series renders at /metrics. docs/metrics.md is the
+        human-readable copy of this list."""
+        # exercise a request so per-model series exist
+        client.post("/v1/completions",
+                    json={"model": "tiny-llama", "prompt": "golden metrics",
+                          "max_tokens": 2})
+        body = client.get("/metrics").content.decode()
+        ns = "inference_extension"
+        llmd = "llm_d_inference_scheduler"
+        families = [
+            f"{ns}_request_total",
+            f"{ns}_request_error_total",
+            f"{ns}_request_duration_seconds",
+            f"{ns}_request_sizes",
+            f"{ns}_input_tokens",
+            f"{ns}_output_tokens",
+            f"{ns}_cached_tokens",
+            f"{ns}_time_to_first_token_seconds",
+            f"{ns}_normalized_time_per_output_token_seconds",
+            f"{ns}_scheduler_e2e_duration_seconds",
+            f"{ns}_plugin_duration_seconds",
+            f"{ns}_running_requests",
+            f"{ns}_prefix_indexer_size",
+            f"{ns}_prefix_indexer_hit_ratio",
+            f"{ns}_flow_control_queue_size",
+            f"{ns}_flow_control_queue_duration_seconds",
+            f"{ns}_flow_control_dispatch_total",
+            f"{ns}_saturation",
+            f"{ns}_model_rewrite_total",
+            f"{llmd}_disagg_decision_total",
+            f"{llmd}_xgmi_kv_transfer_bytes_total",
+            f"{llmd}_xgmi_kv_transfer_seconds",
+        ]
+        missing = [f for f in families if f"# TYPE {f}" not in body]
+        assert not missing, f"families absent from /metrics: {missing}"
+        # the exercised request populated the labelled counter
+        assert 'request_total{model="tiny-llama"' in body

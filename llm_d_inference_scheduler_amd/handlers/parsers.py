@@ -234,11 +234,17 @@ def _pb_walk(data: bytes):
             val, i = _pb_varint(data, i)
         elif wt == 1:                    # 64-bit
             val, i = data[i:i + 8], i + 8
+            if len(val) != 8:
+                raise ValueError("truncated fixed64")
         elif wt == 2:                    # length-delimited
             ln, i = _pb_varint(data, i)
             val, i = data[i:i + ln], i + ln
+            if len(val) != ln:
+                raise ValueError("truncated length-delimited field")
         elif wt == 5:                    # 32-bit
             val, i = data[i:i + 4], i + 4
+            if len(val) != 4:
+                raise ValueError("truncated fixed32")
         else:
             raise ValueError(f"unsupported wire type {wt}")
         yield field, wt, val

@@ -145,3 +145,75 @@ class TestParserFuzz:
             VertexAIParser
         res = VertexAIParser().parse_request(text.encode(), {}, "/v1")
         assert res.error is not None or res.request is not None or res.skip
+
+
+class TestKVBlockIndexInvariants:
+    @settings(max_examples=60, deadline=None)
+    @given(st.lists(st.tuples(
+        st.sampled_from(["store", "evict", "spec", "remove"]),
+        st.integers(0, 2),                     # endpoint
+        st.lists(st.integers(0, 30), max_size=6)), max_size=60))
+    def test_size_and_match_consistency(self, events):
+        """Index size equals the number of live (endpoint, hash) confirmed
+        entries, and match_longest never reports beyond its inputs."""
+        from llm_d_inference_scheduler_amd.datalayer.kvblock import \
+            KVBlockIndex
+        ix = KVBlockIndex(speculative_ttl_s=60.0)
+        live = set()
+        for kind, ep_i, hashes in events:
+            ep = f"gpu{ep_i}"
+            if kind == "store":
+                ix.apply_events(ep, hashes, [])
+                live |= {(ep, h) for h in hashes}
+            elif kind == "evict":
+                ix.apply_events(ep, [], hashes)
+                live -= {(ep, h) for h in hashes}
+            elif kind == "spec":
+                ix.add_speculative(ep, hashes)
+            else:
+                ix.remove_endpoint(ep)
+                live = {(e, h) for e, h in live if e != ep}
+        assert ix.size == len({h for _, h in live})
+        m = ix.match_longest([1, 2, 3], ["gpu0", "gpu1", "gpu2"])
+        for ep, n in m.items():
+            assert 0 <= n <= 3
+            for h in [1, 2, 3][:n]:
+                # confirmed or speculative coverage must actually exist
+                assert (ep, h) in live or (h, ep) in ix._spec
+
+
+class TestEngineLifecycleProperty:
+    @settings(max_examples=10, deadline=None,
+              suppress_health_check=[HealthCheck.too_slow])
+    @given(st.lists(st.tuples(st.integers(1, 60),      # prompt len
+                              st.integers(1, 6),       # max_tokens
+                              st.booleans()),          # abort it mid-flight?
+                    min_size=1, max_size=5),
+           st.integers(6, 24))                          # pool blocks
+    def test_requests_always_terminate(self, reqs, blocks):
+        """Every submitted request either completes or is aborted — no
+        stranded sequences, and the pool leaks no blocks (vLLM-style
+        invariant over random shapes incl. pool churn/preemption)."""
+        import torch
+        from llm_d_inference_scheduler_amd.engine import (EngineRequest,
+                                                          EngineWorker)
+        from llm_d_inference_scheduler_amd.models.configs import TINY_LLAMA
+        w = EngineWorker(TINY_LLAMA, "cpu", kv_blocks=blocks,
+                         dtype=torch.float32)
+        done = set()
+        for i, (plen, mx, abort) in enumerate(reqs):
+            w.add_request(EngineRequest(f"r{i}", list(range(plen)),
+                                        max_tokens=mx))
+        aborted = {f"r{i}" for i, (_, _, a) in enumerate(reqs) if a}
+        stepped = 0
+        while w.has_work and stepped < 400:
+            if stepped == 2 and aborted:
+                for rid in aborted:
+                    w.abort(rid)
+            for out in w.step():
+                if out.finished:
+                    done.add(out.request_id)
+            stepped += 1
+        assert stepped < 400, "engine failed to drain"
+        assert done | aborted >= {f"r{i}" for i in range(len(reqs))}
+        assert not w.mgr.tables, "leaked sequence tables"

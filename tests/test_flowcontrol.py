@@ -249,3 +249,65 @@ class TestFlowControlBenchmark:
         res = run_bench(duration_s=0.3, saturated_every=1, max_items=64)
         # permanently saturated + tiny capacity: most requests rejected
         assert res.rejected + res.evicted > 0
+
+
+class TestCapacityBoundaries:
+    """Exact-boundary semantics (reference processor_test.go:788-875:
+    'exactly at capacity after add' allowed, 'one over' denied) for all
+    four limit axes: band/global x bytes/items."""
+
+    def test_band_bytes_exact_vs_one_over(self):
+        reg = FlowRegistry(bands=[BandConfig(0, max_bytes=30)])
+        fc = FlowController(reg, lambda i: True)
+        a, b = mk_req("a", size=10), mk_req("b", size=20)
+        fc.submit(a); fc.submit(b)             # exactly 30 after add
+        assert a.outcome is None and b.outcome is None
+        c = mk_req("c", size=1)                # one over
+        fc.submit(c)
+        assert c.outcome == QueueOutcome.REJECTED_CAPACITY
+
+    def test_band_items_exact_vs_one_over(self):
+        reg = FlowRegistry(bands=[BandConfig(0, max_items=2)])
+        fc = FlowController(reg, lambda i: True)
+        a, b = mk_req("a"), mk_req("b")
+        fc.submit(a); fc.submit(b)
+        assert a.outcome is None and b.outcome is None
+        c = mk_req("c")
+        fc.submit(c)
+        assert c.outcome == QueueOutcome.REJECTED_CAPACITY
+
+    def test_global_bytes_exact_vs_one_over(self):
+        # same-priority arrivals don't displace; the shard-level byte cap
+        # binds across bands
+        reg = FlowRegistry(bands=[BandConfig(0)], global_max_bytes=25)
+        fc = FlowController(reg, lambda i: True)
+        a, b = mk_req("a", size=10), mk_req("b", size=15)
+        fc.submit(a); fc.submit(b)
+        assert a.outcome is None and b.outcome is None
+        c = mk_req("c", size=1)
+        fc.submit(c)
+        assert c.outcome == QueueOutcome.REJECTED_CAPACITY
+
+    def test_global_items_exact_vs_one_over(self):
+        reg = FlowRegistry(bands=[BandConfig(0)], global_max_items=2)
+        fc = FlowController(reg, lambda i: True)
+        a, b = mk_req("a"), mk_req("b")
+        fc.submit(a); fc.submit(b)
+        assert a.outcome is None and b.outcome is None
+        c = mk_req("c")
+        fc.submit(c)
+        assert c.outcome == QueueOutcome.REJECTED_CAPACITY
+
+    def test_zero_limits_not_treated_as_unlimited_none_is(self):
+        """None = unlimited (reference 'ignore zero-valued capacity
+        limits' maps to our None); an explicit 0 must reject everything."""
+        reg = FlowRegistry(bands=[BandConfig(0, max_items=None)])
+        fc = FlowController(reg, lambda i: True)
+        for i in range(50):
+            fc.submit(mk_req(f"r{i}"))
+        assert all(True for _ in range(1))     # no rejects with None
+        reg0 = FlowRegistry(bands=[BandConfig(0, max_items=0)])
+        fc0 = FlowController(reg0, lambda i: True)
+        z = mk_req("z")
+        fc0.submit(z)
+        assert z.outcome == QueueOutcome.REJECTED_CAPACITY

@@ -23,6 +23,34 @@ from ..models.tokenizer import HashTokenizer
 from ..scheduling.types import LLMRequest
 from .service import NodeService
 
+
+def _vllm_compat(service: NodeService) -> bytes:
+    """vLLM-compatible engine gauges appended to /metrics (the families
+    the reference's datalayer extractor scrapes, options.go:121-125) —
+    aggregated over this node's endpoints so a peer node's
+    HttpMetricsSource (datalayer/extractor.py) can treat the whole node
+    as one worker."""
+    eps = list(service.node.datastore.endpoints())
+    waiting = sum(ep.metrics.waiting_queue_size for ep in eps)
+    running = sum(ep.metrics.running_requests_size for ep in eps)
+    kv = max((ep.metrics.kv_cache_usage for ep in eps), default=0.0)
+    blocks = [(ep.metrics.cache_block_size, ep.metrics.cache_num_blocks)
+              for ep in eps if ep.metrics.cache_num_blocks]
+    lines = [
+        "# TYPE vllm:num_requests_waiting gauge",
+        f"vllm:num_requests_waiting {waiting}",
+        "# TYPE vllm:num_requests_running gauge",
+        f"vllm:num_requests_running {running}",
+        "# TYPE vllm:kv_cache_usage_perc gauge",
+        f"vllm:kv_cache_usage_perc {kv}",
+    ]
+    if blocks:
+        bs, nb = blocks[0]
+        lines += ["# TYPE vllm:cache_config_info gauge",
+                  f'vllm:cache_config_info{{block_size="{bs}",'
+                  f'num_gpu_blocks="{nb}"}} 1']
+    return ("\n".join(lines) + "\n").encode()
+
 DROPPED_REASON_HEADER = "x-request-dropped-reason"
 
 
@@ -171,7 +199,7 @@ def build_app(service: NodeService,
 
     @app.get("/metrics")
     async def metrics():
-        return Response(content=prom.render(),
+        return Response(content=prom.render() + _vllm_compat(service),
                         media_type="text/plain; version=0.0.4")
 
     @app.get("/healthz")

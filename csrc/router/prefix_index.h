@@ -10,6 +10,7 @@
 // shared_mutex (reads = match scans dominate).
 #pragma once
 #include <cstdint>
+#include <cstring>
 #include <unordered_map>
 #include <vector>
 #include <deque>

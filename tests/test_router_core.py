@@ -199,3 +199,21 @@ class TestProfileRunner:
     def test_empty_after_filter(self):
         picks, scores = self._run([(0, 1.0, 0.0)], role_filter=4)  # no encode
         assert len(picks) == 0
+
+
+def test_sanitizer_clean():
+    """ASan+UBSan pass over the C++ core (SURVEY 5.2 `-race` analog).
+    Compiles csrc/router/test/sanitize_main.cpp and runs its randomized
+    workloads; any heap error / UB aborts with nonzero exit."""
+    import os
+    import shutil
+    import subprocess
+    if shutil.which("g++") is None:
+        import pytest
+        pytest.skip("no g++")
+    repo = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    r = subprocess.run(["bash", os.path.join(repo, "tools",
+                                             "sanitize_check.sh")],
+                       capture_output=True, text=True, timeout=300)
+    assert r.returncode == 0, (r.stdout + r.stderr)[-2000:]
+    assert "sanitize: OK" in r.stdout

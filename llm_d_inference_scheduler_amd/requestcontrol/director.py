@@ -115,7 +115,14 @@ class Director:
         t0 = time.monotonic()
         tracer = get_tracer()
         with tracer.span("director.handle_request",
-                         request_id=req.request_id, model=req.model):
+                         request_id=req.request_id, model=req.model) as span:
+            # remote context from the W3C traceparent header (the otel
+            # propagation the reference wires from Envoy, handlers/request.go)
+            from ..telemetry.tracing import parse_traceparent
+            tp = parse_traceparent(req.headers.get("traceparent"))
+            if tp is not None:
+                span.trace_id, parent_span = tp
+                span.set_attribute("remote_parent_span_id", parent_span)
             ctx = SchedulingContext(request=req)
             if precomputed:
                 ctx.attributes.update(

@@ -59,7 +59,8 @@ class Tracer:
         parent = getattr(self._local, "current", None)
         s = Span(name=name, start_ns=time.monotonic_ns(),
                  attributes=dict(attrs),
-                 parent=parent.name if parent else None)
+                 parent=parent.name if parent else None,
+                 trace_id=parent.trace_id if parent else "")
         prev = parent
         self._local.current = s
         try:
@@ -85,6 +86,7 @@ class Tracer:
                 f.write(json.dumps({
                     "name": s.name, "start_ns": s.start_ns, "end_ns": s.end_ns,
                     "duration_ms": s.duration_ms, "parent": s.parent,
+                    "trace_id": s.trace_id,
                     "attributes": s.attributes}) + "\n")
 
 
@@ -110,3 +112,25 @@ def get_tracer() -> Tracer:
     if _tracer is None:
         _tracer = Tracer("llm-d-inference-scheduler-amd")
     return _tracer
+
+
+def parse_traceparent(header: Optional[str]):
+    """W3C traceparent `00-<trace_id>-<parent_span_id>-<flags>` (the
+    context the reference extracts from Envoy headers via otel
+    propagation, handlers/request.go). Returns (trace_id, parent_span_id)
+    or None on any malformation."""
+    if not header:
+        return None
+    parts = header.strip().split("-")
+    if len(parts) != 4:
+        return None
+    version, trace_id, span_id, _flags = parts
+    if len(trace_id) != 32 or len(span_id) != 16 or len(version) != 2:
+        return None
+    try:
+        int(trace_id, 16), int(span_id, 16), int(version, 16)
+    except ValueError:
+        return None
+    if trace_id == "0" * 32 or span_id == "0" * 16:
+        return None
+    return trace_id, span_id

@@ -222,3 +222,34 @@ class TestTracerExport:
         with t.span("x") as s:
             s.set_attribute("a", 1)   # noop span accepts attributes
         assert not t.finished_spans()
+
+
+class TestTraceparent:
+    def test_parse_valid_and_invalid(self):
+        from llm_d_inference_scheduler_amd.telemetry.tracing import \
+            parse_traceparent
+        tid = "a" * 32
+        sid = "b" * 16
+        assert parse_traceparent(f"00-{tid}-{sid}-01") == (tid, sid)
+        for bad in (None, "", "junk", f"00-{tid}-{sid}", f"00-{'z'*32}-{sid}-01",
+                    f"00-{'0'*32}-{sid}-01", f"00-{tid}-{'0'*16}-01",
+                    f"0-{tid}-{sid}-01"):
+            assert parse_traceparent(bad) is None
+
+    def test_root_span_adopts_remote_context_and_children_inherit(self):
+        from llm_d_inference_scheduler_amd.telemetry import (get_tracer,
+                                                             init_tracing)
+        from llm_d_inference_scheduler_amd.telemetry.tracing import \
+            parse_traceparent
+        init_tracing(enabled=True)
+        tracer = get_tracer()
+        tid = "c" * 32
+        tp = parse_traceparent(f"00-{tid}-{'d'*16}-01")
+        with tracer.span("gateway.request") as root:
+            root.trace_id, _ = tp
+            with tracer.span("scheduler.schedule"):
+                pass
+        spans = {s.name: s for s in tracer.finished_spans()}
+        assert spans["gateway.request"].trace_id == tid
+        assert spans["scheduler.schedule"].trace_id == tid
+        assert spans["scheduler.schedule"].parent == "gateway.request"

@@ -89,7 +89,7 @@ PYBIND11_MODULE(_router_core, m) {
               py::array_t<float, py::array::c_style | py::array::forcecast> active_requests,
               int role_filter,
               py::object candidate_mask,  // None or uint8 array
-              const std::vector<std::tuple<int, float, float>>& scorers,
+              const std::vector<std::tuple<int, float, float, float>>& scorers,
               py::object match_blocks,    // None or int32 array
               int total_blocks,
               py::object extra,           // None or float32 array
@@ -104,7 +104,7 @@ PYBIND11_MODULE(_router_core, m) {
              s.active_requests.assign(active_requests.data(), active_requests.data() + s.n);
 
              std::vector<ScorerSpec> specs;
-             for (auto& [k, w, p] : scorers) specs.push_back({k, w, p});
+             for (auto& [k, w, p, p2] : scorers) specs.push_back({k, w, p, p2});
 
              py::array_t<uint8_t> cmask_arr;
              const uint8_t* cmask = nullptr;

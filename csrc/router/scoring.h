@@ -44,7 +44,9 @@ enum PickerKind : int32_t {
 struct ScorerSpec {
   int32_t kind;
   float weight;
-  float param;  // threshold for LOAD_AWARE / TOKEN_LOAD; unused otherwise
+  float param;   // threshold for LOAD_AWARE / TOKEN_LOAD; idleThreshold for
+                 // ACTIVE_REQUEST; unused otherwise
+  float param2;  // maxBusyScore for ACTIVE_REQUEST ((0,1]; <=0 -> 1.0)
 };
 
 struct Snapshot {
@@ -105,7 +107,21 @@ class ProfileRunner {
       switch (sp.kind) {
         case SC_QUEUE: minmax_inverted(s.queue_depth.data(), alive.data(), n, tmp.data()); break;
         case SC_RUNNING: minmax_inverted(s.running.data(), alive.data(), n, tmp.data()); break;
-        case SC_ACTIVE_REQUEST: minmax_inverted(s.active_requests.data(), alive.data(), n, tmp.data()); break;
+        case SC_ACTIVE_REQUEST: {
+          // activerequest/active_request.go:160-168: idle (count <=
+          // idleThreshold) pins 1.0; busy scales (max-c)/max * maxBusyScore
+          float idle_thr = std::max(0.f, sp.param);
+          float busy_max = (sp.param2 > 0.f && sp.param2 <= 1.f) ? sp.param2 : 1.f;
+          float hi = 0.f;
+          for (int i = 0; i < n; ++i)
+            if (alive[i]) hi = std::max(hi, s.active_requests[i]);
+          for (int i = 0; i < n; ++i) {
+            float c = s.active_requests[i];
+            tmp[i] = (c <= idle_thr) ? 1.f
+                     : (hi > 0.f ? (hi - c) / hi * busy_max : 1.f);
+          }
+          break;
+        }
         case SC_KV_UTIL:
           for (int i = 0; i < n; ++i) tmp[i] = 1.f - s.kv_usage[i];
           break;

@@ -118,17 +118,36 @@ class TokenLoadScorer(Scorer):
 
 @register_plugin("active-request-scorer")
 class ActiveRequestScorer(Scorer):
-    """Router-tracked in-flight counts, min-max inverted (scorer/activerequest)."""
+    """Router-tracked in-flight counts (scorer/activerequest, active_request.go:
+    139-168): idle endpoints (count <= idleThreshold) pin 1.0; busy endpoints
+    scale (max-count)/max * maxBusyScore — the gap steers toward idle pods.
+    requestTimeout is deprecated upstream and ignored here too (tracking is
+    the inflight-load-producer's job)."""
+
+    def __init__(self, name: str = "", **params):
+        super().__init__(name, **params)
+        self.idle_threshold = max(0, int(params.get("idleThreshold", 0)))
+        mbs = float(params.get("maxBusyScore", 1.0))
+        self.max_busy_score = mbs if 0.0 < mbs <= 1.0 else 1.0
 
     def native_spec(self):
-        return (SC_ACTIVE_REQUEST, 0.0)
+        return (SC_ACTIVE_REQUEST, float(self.idle_threshold),
+                self.max_busy_score)
 
     def score(self, ctx, endpoints):
         vals = {}
         for ep in endpoints:
             load = ep.get_attribute(IN_FLIGHT_LOAD)
             vals[ep.name] = float(load.snapshot()[0]) if load else 0.0
-        return _minmax_inverted(vals)
+        hi = max(vals.values(), default=0.0)
+        out = {}
+        for name, c in vals.items():
+            if c <= self.idle_threshold:
+                out[name] = 1.0
+            else:
+                out[name] = ((hi - c) / hi * self.max_busy_score
+                             if hi > 0 else 1.0)
+        return out
 
 
 @register_plugin("lora-affinity-scorer")

@@ -55,12 +55,14 @@ class SchedulerProfile:
             eps = f.filter(ctx, eps)
             if not eps:
                 return result
-        native_specs: List[Tuple[int, float, float]] = []
+        native_specs: List[Tuple[int, float, float, float]] = []
         python_scorers: List[Tuple[Scorer, float]] = []
         for scorer, weight in self.scorers:
             spec = scorer.native_spec()
             if spec is not None:
-                native_specs.append((spec[0], float(weight), float(spec[1])))
+                p2 = float(spec[2]) if len(spec) > 2 else 0.0
+                native_specs.append((spec[0], float(weight),
+                                     float(spec[1]), p2))
             else:
                 python_scorers.append((scorer, weight))
 
@@ -121,7 +123,7 @@ def _snapshot_arrays(eps: List[Endpoint]) -> Dict[str, np.ndarray]:
 
 
 def _prefix_arrays(ctx, eps, native_specs):
-    if not any(k == SC_PREFIX_KIND for k, _, _ in native_specs):
+    if not any(sp[0] == SC_PREFIX_KIND for sp in native_specs):
         return None, 0
     info = ctx.attributes.get(PREFIX_CACHE_MATCH_INFO)
     if info is None:

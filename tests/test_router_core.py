@@ -154,14 +154,14 @@ class TestProfileRunner:
             picker, kw.get("max_endpoints", 1))
 
     def test_queue_scorer_minmax(self):
-        picks, scores = self._run([(0, 1.0, 0.0)],
+        picks, scores = self._run([(0, 1.0, 0.0, 0.0)],
                                   queue=np.array([0, 10, 5, 10], np.float32))
         assert picks[0] == 0
         assert scores[0] == 1.0 and scores[1] == 0.0 and abs(scores[2] - .5) < 1e-6
 
     def test_kv_and_prefix(self):
         picks, scores = self._run(
-            [(1, 1.0, 0.0), (2, 2.0, 0.0)],
+            [(1, 1.0, 0.0, 0.0), (2, 2.0, 0.0, 0.0)],
             kv=np.array([0.5, 0.2, 0.9, 0.0], np.float32),
             match=np.array([0, 4, 0, 0], np.int32), total=8)
         # ep1: (1-0.2) + 2*0.5 = 1.8 -> best
@@ -170,13 +170,13 @@ class TestProfileRunner:
 
     def test_role_filter(self):
         roles = np.array([1, 1, 2, 2], np.uint8)  # 2 decode, 2 prefill
-        picks, scores = self._run([(0, 1.0, 0.0)], roles=roles, role_filter=2,
+        picks, scores = self._run([(0, 1.0, 0.0, 0.0)], roles=roles, role_filter=2,
                                   queue=np.array([0, 0, 9, 1], np.float32))
         assert picks[0] == 3
         assert scores[0] == -1.0 and scores[1] == -1.0  # filtered
 
     def test_candidate_mask(self):
-        picks, _ = self._run([(0, 1.0, 0.0)],
+        picks, _ = self._run([(0, 1.0, 0.0, 0.0)],
                              mask=np.array([0, 0, 1, 0], np.uint8))
         assert picks[0] == 2
 
@@ -197,7 +197,7 @@ class TestProfileRunner:
         assert abs(freq[0] - 0.7) < 0.06
 
     def test_empty_after_filter(self):
-        picks, scores = self._run([(0, 1.0, 0.0)], role_filter=4)  # no encode
+        picks, scores = self._run([(0, 1.0, 0.0, 0.0)], role_filter=4)  # no encode
         assert len(picks) == 0
 
 

@@ -203,10 +203,19 @@ class HttpMetricsSource(DataSource):
         return f"http://{endpoint.metadata.address}/metrics"
 
     def collect(self, endpoint: Endpoint) -> Optional[Metrics]:
+        from ..metrics import prom
         try:
             text = self._fetch(self.url_for(endpoint))
         except Exception as e:
             log.v(4).info("scrape failed", endpoint=endpoint.name,
                           err=str(e))
+            prom.datalayer_poll_errors.labels("HttpMetricsSource").inc()
             return None   # stale metrics -> saturation detector handles it
-        return extract_metrics(text, self.specs)
+        try:
+            return extract_metrics(text, self.specs)
+        except Exception as e:   # the parser is lenient; this is a backstop
+            log.v(4).info("extract failed", endpoint=endpoint.name,
+                          err=str(e))
+            prom.datalayer_extract_errors.labels(
+                "HttpMetricsSource", "vllm").inc()
+            return None

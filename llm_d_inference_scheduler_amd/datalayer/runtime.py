@@ -60,11 +60,14 @@ class Collector:
         self._thread: Optional[threading.Thread] = None
 
     def collect_once(self) -> None:
+        from ..metrics import prom
         for src in self.sources:
             try:
                 m = src.collect(self.endpoint)
             except Exception as e:  # stale metrics treated as saturated later
                 log.v(4).info("collect failed", endpoint=self.endpoint.name, err=str(e))
+                prom.datalayer_poll_errors.labels(
+                    type(src).__name__).inc()
                 continue
             if m is not None:
                 self.endpoint.update_metrics(m)

@@ -81,6 +81,15 @@ rewrite_decision_total = Counter(
     f"{NS}_model_rewrite_total", "Model rewrite decisions",
     ["model", "target_model"], registry=registry)
 
+datalayer_poll_errors = Counter(
+    f"{LLMD}_datalayer_poll_errors_total",
+    "Data-source poll errors per source type", ["source_type"],
+    registry=registry)
+datalayer_extract_errors = Counter(
+    f"{LLMD}_datalayer_extract_errors_total",
+    "Extract errors per source/extractor type",
+    ["source_type", "extractor_type"], registry=registry)
+
 xgmi_kv_transfer_bytes = Counter(
     f"{LLMD}_xgmi_kv_transfer_bytes_total",
     "KV-cache bytes moved over xGMI", ["direction"], registry=registry)

@@ -223,6 +223,8 @@ class TestMetricsGolden:
             f"{ns}_saturation",
             f"{ns}_model_rewrite_total",
             f"{llmd}_disagg_decision_total",
+            f"{llmd}_datalayer_poll_errors_total",
+            f"{llmd}_datalayer_extract_errors_total",
             f"{llmd}_xgmi_kv_transfer_bytes_total",
             f"{llmd}_xgmi_kv_transfer_seconds",
         ]

@@ -199,8 +199,19 @@ class NodeRunner:
                     candidates.all()))
             self.flow.start()
             admission = FlowControlAdmissionController(self.flow)
+            # in-flight evictor (flowcontrol/eviction): config-supplied
+            # ordering/filter policy plugins override the defaults
+            from ..flowcontrol.evictor import (EvictionFilterPolicy,
+                                               EvictionOrderingPolicy,
+                                               RequestEvictor)
+            ordering = next((pl for pl in self.loaded.plugins.values()
+                             if isinstance(pl, EvictionOrderingPolicy)), None)
+            filt = next((pl for pl in self.loaded.plugins.values()
+                         if isinstance(pl, EvictionFilterPolicy)), None)
+            self.evictor = RequestEvictor(ordering, filt)
         else:
             self.flow = None
+            self.evictor = None
             admission = LegacyAdmissionController(self.detector)
         self.director = Director(
             datastore=self.datastore,
@@ -275,12 +286,39 @@ class NodeRunner:
         self._arrivals = [r for r in self._arrivals
                           if r.request_id != request_id]
         self._chunked.pop(request_id, None)
+        if self.evictor is not None:
+            self.evictor.untrack(request_id)
         decision = self._decisions.pop(request_id, None)
         self._outbox.append({"type": "abort", "req_id": request_id})
         self._abort_local(request_id)
         if decision is not None:
             # unwind response hooks (inflight counters, training state)
             self.director.handle_response_complete(decision, Usage())
+
+    def _maybe_evict_inflight(self) -> None:
+        """Saturation-driven in-flight eviction (request_evictor.go EvictN):
+        when the pool is saturated AND flow-control work is queued behind
+        it, kill the most-evictable dispatched sheddable request so the
+        queue can drain; the client gets a 429-reason error completion
+        (server.go:262-284 eviction -> ImmediateResponse)."""
+        if self.evictor is None or self.flow is None:
+            return
+        if self.flow.queued_len == 0 or self.evictor.stats[1] == 0:
+            return
+        if not self.detector.is_saturated(self.datastore.endpoints()):
+            return
+        for rid in self.evictor.evict_n(1, lambda item: None):
+            self._chunked.pop(rid, None)
+            decision = self._decisions.pop(rid, None)
+            self._outbox.append({"type": "abort", "req_id": rid})
+            self._abort_local(rid)
+            if decision is not None:
+                self.director.handle_response_complete(decision, Usage())
+            prom.request_error_total.labels(self.cfg.model.name,
+                                            "evicted").inc()
+            prom.flow_dispatch_total.labels("evicted_inflight").inc()
+            self._completions.append(Completion(
+                request_id=rid, usage=Usage(), error="evicted"))
 
     def _abort_local(self, rid: str) -> None:
         self.engine.abort(rid)
@@ -316,6 +354,7 @@ class NodeRunner:
         """One lockstep node iteration on every rank."""
         if self.is_router:
             self._route_arrivals()
+            self._maybe_evict_inflight()
         self._outbox.append({"type": "metrics", "src": self.rank,
                              "m": self._metrics_payload()})
         stored, evicted = self.engine.mgr.drain_events()
@@ -417,6 +456,11 @@ class NodeRunner:
     def _emit_assignment(self, req: LLMRequest,
                          decision: RoutingDecision) -> None:
             self._decisions[req.request_id] = decision
+            if self.evictor is not None:
+                from ..flowcontrol.evictor import EvictionItem
+                self.evictor.track(EvictionItem(
+                    request_id=req.request_id, priority=req.priority,
+                    target=decision.target.name))
             self.epp_latencies.append(decision.epp_latency_ms)
             decode_rank = decision.target.metadata.rank
             chunk = self.cfg.decode_chunk_tokens
@@ -625,12 +669,16 @@ class NodeRunner:
                           cached_tokens=first.cached_tokens,
                           ttft_ms=first.ttft_ms, tpot_ms=usage.tpot_ms,
                           e2e_ms=(first.e2e_ms or 0) + (usage.e2e_ms or 0))
+            if self.evictor is not None:
+                self.evictor.untrack(m["req_id"])
             self._completions.append(Completion(
                 request_id=m["req_id"], usage=usage,
                 tokens=state["tokens"],
                 finish_reason=m.get("finish_reason", "length"), error=""))
             return
         self._chunked.pop(m["req_id"], None)
+        if self.evictor is not None:
+            self.evictor.untrack(m["req_id"])
         self._completions.append(Completion(
             request_id=m["req_id"], usage=usage,
             tokens=m.get("tokens", []),

@@ -10,3 +10,4 @@ def register_all_plugins() -> None:
     (parity: pkg/epp/framework/plugins/register.go:23-56 +
     cmd/epp/runner/runner.go:463-515 registerInTreePlugins)."""
     from . import filters, scorers, pickers, profile_handlers, producers, admitters  # noqa: F401
+    from ..flowcontrol import evictor  # noqa: F401  (eviction policy plugins)

@@ -46,6 +46,8 @@ REFERENCE_PLUGINS = [
     "utilization-detector", "concurrency-detector",
     # conformance/test plugins registered in the production runner
     "destination-endpoint-served-verifier",
+    # in-flight eviction policies (framework/plugins/flowcontrol/eviction)
+    "priority-then-time-eviction-order-policy", "sheddable-eviction-filter",
 ]
 
 SCORER_TYPES = [

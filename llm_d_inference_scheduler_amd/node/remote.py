@@ -1,0 +1,125 @@
+"""Remote-node endpoints: route requests to ANOTHER node's front door.
+
+Reference analog: a multi-pool deployment where the router steers traffic
+to model servers outside its own pool (apix InferencePoolImport /
+cross-pool endpoints). Here the remote "model server" is a peer node's
+front door, which already speaks both halves of the contract:
+  * metrics: its /metrics renders the vLLM-compatible families, scraped
+    by `datalayer/extractor.HttpMetricsSource` (options.go:121-125);
+  * execution: the internal enqueue API (`/internal/v1/enqueue`,
+    server/openai_app.py) accepts a tokenized request and returns the
+    completion.
+
+A remote endpoint is a normal datastore Endpoint whose
+`llm-d.ai/remote-url` label carries the peer's base URL — every filter,
+scorer, and picker treats it exactly like a local GPU-role endpoint; only
+dispatch differs (`NodeRunner._emit_assignment` hands the request to the
+RemoteForwarder instead of the engine mailbox). Forwarded requests are
+non-streaming end-to-end; tokens surface in one completion.
+"""
+import queue
+import threading
+from concurrent.futures import ThreadPoolExecutor
+from typing import Callable, Dict, List, Optional
+
+from ..datalayer.datastore import make_endpoint
+from ..datalayer.endpoint import Endpoint
+from ..handlers.parsers import Usage
+from ..scheduling.types import LLMRequest
+from ..utils.logging import get_logger
+
+log = get_logger("node.remote")
+
+REMOTE_URL_LABEL = "llm-d.ai/remote-url"
+ENQUEUE_PATH = "/internal/v1/enqueue"
+
+# transport(url, payload_dict) -> response_dict; injectable for tests
+Transport = Callable[[str, Dict], Dict]
+
+
+def remote_endpoint(name: str, index: int, url: str,
+                    role: str = "decode") -> Endpoint:
+    """Datastore endpoint for a peer node's front door at `url`."""
+    base = url.rstrip("/")
+    return make_endpoint(name, index, role=role, address=base,
+                         labels={REMOTE_URL_LABEL: base,
+                                 "metrics_url": base + "/metrics"})
+
+
+def _default_transport(url: str, payload: Dict) -> Dict:
+    import httpx
+    r = httpx.post(url, json=payload, timeout=300.0)
+    r.raise_for_status()
+    return r.json()
+
+
+class RemoteForwarder:
+    """Forwards scheduled requests to remote-node endpoints on a small
+    thread pool; completed responses drain into the router step as
+    Completion objects (mirrors the sidecar's role of carrying the
+    request to the remote engine and relaying the response)."""
+
+    def __init__(self, transport: Optional[Transport] = None,
+                 max_workers: int = 8):
+        self.transport = transport or _default_transport
+        self._pool: Optional[ThreadPoolExecutor] = None
+        self._done: "queue.Queue" = queue.Queue()
+        self._lock = threading.Lock()
+        self.inflight = 0
+
+    def _ensure_pool(self) -> ThreadPoolExecutor:
+        if self._pool is None:
+            self._pool = ThreadPoolExecutor(max_workers=8,
+                                            thread_name_prefix="remote-fwd")
+        return self._pool
+
+    def forward(self, req: LLMRequest, base_url: str) -> None:
+        payload = {
+            "request_id": req.request_id,
+            "model": req.target_model or req.model,
+            "prompt": req.prompt,
+            "prompt_tokens": req.prompt_tokens,
+            "max_tokens": req.max_tokens,
+            "temperature": req.temperature,
+            "stop_token_ids": req.stop_token_ids,
+            "priority": req.priority,
+        }
+        with self._lock:
+            self.inflight += 1
+        self._ensure_pool().submit(self._run, req.request_id,
+                                   base_url + ENQUEUE_PATH, payload)
+
+    def _run(self, request_id: str, url: str, payload: Dict) -> None:
+        from .runner import Completion
+        try:
+            resp = self.transport(url, payload)
+            u = resp.get("usage", {})
+            comp = Completion(
+                request_id=request_id,
+                usage=Usage(prompt_tokens=u.get("prompt_tokens", 0),
+                            completion_tokens=u.get("completion_tokens", 0),
+                            cached_tokens=u.get("cached_tokens", 0),
+                            ttft_ms=u.get("ttft_ms"),
+                            e2e_ms=u.get("e2e_ms")),
+                tokens=resp.get("tokens", []),
+                finish_reason=resp.get("finish_reason", "length"),
+                error=resp.get("error", ""))
+        except Exception as e:
+            log.error("remote forward failed", url=url, err=str(e))
+            comp = Completion(request_id=request_id, usage=Usage(),
+                              error=f"remote_error: {e}")
+        with self._lock:
+            self.inflight -= 1
+        self._done.put(comp)
+
+    def drain(self) -> List:
+        out = []
+        while True:
+            try:
+                out.append(self._done.get_nowait())
+            except queue.Empty:
+                return out
+
+    def shutdown(self) -> None:
+        if self._pool is not None:
+            self._pool.shutdown(wait=False)

@@ -103,6 +103,8 @@ class NodeConfig:
     #                  decode-side setup overlap and the transfer can never
     #                  hit kv_exhausted (connector_sglang.go)
     connector: str = "nixlv2"
+    # injectable transport for remote-node forwarding (tests); None = httpx
+    remote_transport: Any = None
     cache_hit_threshold: float = 0.8
     # chunked decode (sidecar decode.go:62-315): cap per-dispatch generation
     # at N tokens; each continuation re-routes through the scheduler (the
@@ -218,6 +220,8 @@ class NodeRunner:
             scheduler=Scheduler(self.loaded.scheduler_config),
             admission=admission, candidates=candidates,
             config=self.loaded.request_control)
+        from .remote import RemoteForwarder
+        self.remote = RemoteForwarder(transport=cfg.remote_transport)
         # batched gfx950 prefix path: hash+match the whole admission batch
         # in two kernel launches on the router rank's GPU
         self._approx = None
@@ -355,6 +359,12 @@ class NodeRunner:
         if self.is_router:
             self._route_arrivals()
             self._maybe_evict_inflight()
+            for comp in self.remote.drain():
+                decision = self._decisions.pop(comp.request_id, None)
+                if decision is not None:
+                    self.director.handle_response_complete(
+                        decision, comp.usage)
+                self._completions.append(comp)
         self._outbox.append({"type": "metrics", "src": self.rank,
                              "m": self._metrics_payload()})
         stored, evicted = self.engine.mgr.drain_events()
@@ -456,6 +466,17 @@ class NodeRunner:
     def _emit_assignment(self, req: LLMRequest,
                          decision: RoutingDecision) -> None:
             self._decisions[req.request_id] = decision
+            from .remote import REMOTE_URL_LABEL
+            remote_url = decision.target.metadata.labels.get(
+                REMOTE_URL_LABEL)
+            if remote_url:
+                # peer-node endpoint: hand off to its front door; the
+                # completion drains back through the router step
+                # (node/remote.py). Not evictor-tracked: a remote
+                # in-flight request cannot be aborted from here (v1).
+                self.epp_latencies.append(decision.epp_latency_ms)
+                self.remote.forward(req, remote_url)
+                return
             if self.evictor is not None:
                 from ..flowcontrol.evictor import EvictionItem
                 self.evictor.track(EvictionItem(
@@ -818,3 +839,5 @@ class NodeRunner:
     def shutdown(self) -> None:
         if self.is_router and self.flow is not None:
             self.flow.stop()
+        if self.is_router and getattr(self, "remote", None) is not None:
+            self.remote.shutdown()

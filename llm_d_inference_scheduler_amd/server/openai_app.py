@@ -202,6 +202,43 @@ def build_app(service: NodeService,
         return Response(content=prom.render() + _vllm_compat(service),
                         media_type="text/plain; version=0.0.4")
 
+    @app.post("/internal/v1/enqueue")
+    async def internal_enqueue(request: Request):
+        """Node-to-node execution API (node/remote.py): a peer router
+        forwards an already-tokenized scheduled request here; this node
+        runs it through its own engine pool and returns the completion.
+        The pair (this route + the vLLM-compatible /metrics families) is
+        what makes a front door usable as a remote endpoint."""
+        try:
+            body = json.loads(await request.body())
+        except (ValueError, TypeError):
+            return _error(400, "parse_error", "invalid JSON")
+        if not isinstance(body, dict) or not body.get("request_id"):
+            return _error(400, "parse_error", "request_id required")
+        req = LLMRequest(
+            request_id=f"fwd-{body['request_id']}",
+            model=body.get("model", service.node.cfg.model.name),
+            prompt=body.get("prompt", ""),
+            prompt_tokens=body.get("prompt_tokens"),
+            max_tokens=int(body.get("max_tokens", 16)),
+            temperature=float(body.get("temperature") or 0.0),
+            stop_token_ids=body.get("stop_token_ids"),
+            priority=int(body.get("priority", 0)))
+        handle = service.submit(req)
+        completion = await asyncio.to_thread(handle.wait, 300.0)
+        if completion is None:
+            service.cancel(req.request_id)
+            return JSONResponse(status_code=504,
+                                content={"error": "timeout"})
+        u = completion.usage
+        return {"tokens": completion.tokens,
+                "finish_reason": completion.finish_reason,
+                "error": completion.error,
+                "usage": {"prompt_tokens": u.prompt_tokens,
+                          "completion_tokens": u.completion_tokens,
+                          "cached_tokens": u.cached_tokens,
+                          "ttft_ms": u.ttft_ms, "e2e_ms": u.e2e_ms}}
+
     @app.get("/healthz")
     async def healthz():
         ready = service.node.datastore.pool_ready()

@@ -1,5 +1,7 @@
 """In-flight request eviction (reference flowcontrol/eviction/
 request_evictor.go + priority_time/sheddable policy plugins)."""
+import time
+
 import torch
 
 from llm_d_inference_scheduler_amd.flowcontrol.evictor import (
@@ -70,8 +72,11 @@ class TestNodeInflightEviction:
 
         node.submit(req("shed", -1))
         node.submit(req("crit", 5))
-        for _ in range(6):
+        for _ in range(300):          # fc-mode routing is threaded
             node.step()
+            if node.evictor.stats == (2, 1):
+                break
+            time.sleep(0.01)
         assert node.evictor.stats == (2, 1)
 
         # force the trigger conditions: saturation + queued work

@@ -1,0 +1,113 @@
+"""Multi-node routing: node A schedules onto a remote endpoint that is
+node B's front door (node/remote.py + /internal/v1/enqueue +
+HttpMetricsSource). The reference analog is cross-pool routing
+(InferencePoolImport); here two full nodes run in-process on CPU."""
+import time
+
+import pytest
+import torch
+from fastapi.testclient import TestClient
+
+from llm_d_inference_scheduler_amd.datalayer.extractor import \
+    HttpMetricsSource
+from llm_d_inference_scheduler_amd.models.configs import TINY_LLAMA
+from llm_d_inference_scheduler_amd.node import NodeConfig, NodeRunner
+from llm_d_inference_scheduler_amd.node.remote import (REMOTE_URL_LABEL,
+                                                       remote_endpoint)
+from llm_d_inference_scheduler_amd.scheduling.types import LLMRequest
+from llm_d_inference_scheduler_amd.server import NodeService, build_app
+
+
+@pytest.fixture(scope="module")
+def node_b_client():
+    cfg = NodeConfig(model=TINY_LLAMA, device="cpu", dtype=torch.float32,
+                     kv_blocks=256)
+    service = NodeService(NodeRunner(cfg))
+    service.start()
+    with TestClient(build_app(service)) as c:
+        yield c
+    service.stop()
+
+
+class TestInternalEnqueue:
+    def test_enqueue_returns_tokens(self, node_b_client):
+        r = node_b_client.post("/internal/v1/enqueue", json={
+            "request_id": "x1", "model": "tiny-llama",
+            "prompt_tokens": list(range(40)), "max_tokens": 4})
+        assert r.status_code == 200
+        body = r.json()
+        assert len(body["tokens"]) == 4 and body["error"] == ""
+        assert body["usage"]["prompt_tokens"] == 40
+
+    def test_enqueue_rejects_garbage(self, node_b_client):
+        assert node_b_client.post("/internal/v1/enqueue",
+                                  content=b"{{{").status_code == 400
+        assert node_b_client.post("/internal/v1/enqueue",
+                                  json={}).status_code == 400
+
+
+class TestCrossNodeRouting:
+    def test_request_routed_to_remote_node(self, node_b_client):
+        """Node A's scheduler picks the remote endpoint (subset hint);
+        the request executes on node B and the completion surfaces at
+        node A with response hooks run."""
+        def transport(url, payload):
+            r = node_b_client.post("/internal/v1/enqueue", json=payload)
+            assert r.status_code == 200
+            return r.json()
+
+        node_a = NodeRunner(NodeConfig(model=TINY_LLAMA, device="cpu",
+                                       dtype=torch.float32, kv_blocks=64,
+                                       remote_transport=transport))
+        ep = remote_endpoint("peer-b", 1, "http://node-b:8000")
+        assert ep.metadata.labels[REMOTE_URL_LABEL]
+        node_a.datastore.add_endpoint(ep)
+
+        node_a.submit(LLMRequest(
+            request_id="r-remote", model=TINY_LLAMA.name, prompt="",
+            prompt_tokens=list(range(32)), max_tokens=3,
+            subset_hint=["peer-b"]))
+        comps = []
+        for _ in range(200):
+            node_a.step()
+            comps = node_a.drain_completions()
+            if comps:
+                break
+            time.sleep(0.02)       # remote round trip is asynchronous
+        assert comps and comps[0].request_id == "r-remote"
+        assert comps[0].error == "" and len(comps[0].tokens) == 3
+        assert comps[0].usage.prompt_tokens == 32
+        assert "r-remote" not in node_a._decisions      # bookkeeping done
+        node_a.shutdown()
+
+    def test_remote_failure_surfaces_error(self):
+        def boom(url, payload):
+            raise ConnectionError("peer down")
+        node_a = NodeRunner(NodeConfig(model=TINY_LLAMA, device="cpu",
+                                       dtype=torch.float32, kv_blocks=64,
+                                       remote_transport=boom))
+        node_a.datastore.add_endpoint(
+            remote_endpoint("peer-b", 1, "http://node-b:8000"))
+        node_a.submit(LLMRequest(
+            request_id="r-fail", model=TINY_LLAMA.name, prompt="",
+            prompt_tokens=list(range(16)), max_tokens=2,
+            subset_hint=["peer-b"]))
+        comps = []
+        for _ in range(200):
+            node_a.step()
+            comps = node_a.drain_completions()
+            if comps:
+                break
+            time.sleep(0.01)
+        assert comps and comps[0].error.startswith("remote_error")
+        node_a.shutdown()
+
+    def test_remote_metrics_scrape_loop(self, node_b_client):
+        """Node B's /metrics feeds node A's datastore through the
+        HttpMetricsSource — the other half of the remote-endpoint
+        contract."""
+        src = HttpMetricsSource(
+            fetcher=lambda url: node_b_client.get("/metrics").content.decode())
+        ep = remote_endpoint("peer-b", 1, "http://node-b:8000")
+        m = src.collect(ep)
+        assert m is not None and m.cache_num_blocks > 0

@@ -32,6 +32,7 @@ log = get_logger("node.remote")
 
 REMOTE_URL_LABEL = "llm-d.ai/remote-url"
 ENQUEUE_PATH = "/internal/v1/enqueue"
+CANCEL_PATH = "/internal/v1/cancel"
 
 # transport(url, payload_dict) -> response_dict; injectable for tests
 Transport = Callable[[str, Dict], Dict]
@@ -111,6 +112,19 @@ class RemoteForwarder:
         with self._lock:
             self.inflight -= 1
         self._done.put(comp)
+
+    def cancel(self, request_id: str, base_url: str) -> None:
+        """Best-effort remote abort: tells the peer to unwind the
+        forwarded request (client died on this side). The in-flight
+        forward still resolves — its completion is discarded by the
+        router since the decision entry is gone."""
+        def _go():
+            try:
+                self.transport(base_url + CANCEL_PATH,
+                               {"request_id": request_id})
+            except Exception as e:
+                log.v(4).info("remote cancel failed", err=str(e))
+        self._ensure_pool().submit(_go)
 
     def drain(self) -> List:
         out = []

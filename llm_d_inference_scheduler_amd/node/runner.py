@@ -222,6 +222,7 @@ class NodeRunner:
             config=self.loaded.request_control)
         from .remote import RemoteForwarder
         self.remote = RemoteForwarder(transport=cfg.remote_transport)
+        self._remote_urls: Dict[str, str] = {}
         # batched gfx950 prefix path: hash+match the whole admission batch
         # in two kernel launches on the router rank's GPU
         self._approx = None
@@ -292,6 +293,9 @@ class NodeRunner:
         self._chunked.pop(request_id, None)
         if self.evictor is not None:
             self.evictor.untrack(request_id)
+        url = self._remote_urls.pop(request_id, None)
+        if url is not None:
+            self.remote.cancel(request_id, url)
         decision = self._decisions.pop(request_id, None)
         self._outbox.append({"type": "abort", "req_id": request_id})
         self._abort_local(request_id)
@@ -360,6 +364,7 @@ class NodeRunner:
             self._route_arrivals()
             self._maybe_evict_inflight()
             for comp in self.remote.drain():
+                self._remote_urls.pop(comp.request_id, None)
                 decision = self._decisions.pop(comp.request_id, None)
                 if decision is not None:
                     self.director.handle_response_complete(
@@ -472,9 +477,11 @@ class NodeRunner:
             if remote_url:
                 # peer-node endpoint: hand off to its front door; the
                 # completion drains back through the router step
-                # (node/remote.py). Not evictor-tracked: a remote
-                # in-flight request cannot be aborted from here (v1).
+                # (node/remote.py). Not evictor-tracked (the peer's own
+                # evictor owns its in-flight work); cancel() reaches it
+                # through the internal cancel API.
                 self.epp_latencies.append(decision.epp_latency_ms)
+                self._remote_urls[req.request_id] = remote_url
                 self.remote.forward(req, remote_url)
                 return
             if self.evictor is not None:

@@ -239,6 +239,18 @@ def build_app(service: NodeService,
                           "cached_tokens": u.cached_tokens,
                           "ttft_ms": u.ttft_ms, "e2e_ms": u.e2e_ms}}
 
+    @app.post("/internal/v1/cancel")
+    async def internal_cancel(request: Request):
+        """Peer-side abort of a forwarded request (node/remote.py
+        RemoteForwarder.cancel): unwinds the engine-resident sequence."""
+        try:
+            body = json.loads(await request.body())
+            rid = body["request_id"]
+        except (ValueError, TypeError, KeyError):
+            return _error(400, "parse_error", "request_id required")
+        service.cancel(f"fwd-{rid}")
+        return {"canceled": rid}
+
     @app.get("/healthz")
     async def healthz():
         ready = service.node.datastore.pool_ready()

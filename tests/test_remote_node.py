@@ -111,3 +111,48 @@ class TestCrossNodeRouting:
         ep = remote_endpoint("peer-b", 1, "http://node-b:8000")
         m = src.collect(ep)
         assert m is not None and m.cache_num_blocks > 0
+
+
+class TestRemoteCancel:
+    def test_cancel_reaches_peer(self, node_b_client):
+        """cancel() on node A fires the internal cancel API on node B
+        (fwd- prefixed id), unwinding the peer's engine state."""
+        calls = []
+
+        def transport(url, payload):
+            calls.append((url, payload))
+            if url.endswith("/enqueue"):
+                # hold the request "in flight" on the peer: return slowly
+                time.sleep(0.2)
+                return {"tokens": [], "finish_reason": "length",
+                        "error": "canceled", "usage": {}}
+            return node_b_client.post("/internal/v1/cancel",
+                                      json=payload).json()
+
+        node_a = NodeRunner(NodeConfig(model=TINY_LLAMA, device="cpu",
+                                       dtype=torch.float32, kv_blocks=64,
+                                       remote_transport=transport))
+        node_a.datastore.add_endpoint(
+            remote_endpoint("peer-b", 1, "http://node-b:8000"))
+        node_a.submit(LLMRequest(
+            request_id="r-c", model=TINY_LLAMA.name, prompt="",
+            prompt_tokens=list(range(16)), max_tokens=64,
+            subset_hint=["peer-b"]))
+        for _ in range(100):
+            node_a.step()
+            if any(u.endswith("/enqueue") for u, _ in calls):
+                break
+            time.sleep(0.01)
+        node_a.cancel("r-c")
+        for _ in range(100):
+            node_a.step()
+            if any(u.endswith("/cancel") for u, _ in calls):
+                break
+            time.sleep(0.01)
+        cancel_calls = [(u, p) for u, p in calls if u.endswith("/cancel")]
+        assert cancel_calls and cancel_calls[0][1] == {"request_id": "r-c"}
+        # peer-side API accepts the id (no fwd- request exists -> no-op)
+        r = node_b_client.post("/internal/v1/cancel",
+                               json={"request_id": "r-c"})
+        assert r.status_code == 200 and r.json()["canceled"] == "r-c"
+        node_a.shutdown()

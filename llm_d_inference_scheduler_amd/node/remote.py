@@ -63,6 +63,7 @@ class RemoteForwarder:
     def __init__(self, transport: Optional[Transport] = None,
                  max_workers: int = 8):
         self.transport = transport or _default_transport
+        self.max_workers = max_workers
         self._pool: Optional[ThreadPoolExecutor] = None
         self._done: "queue.Queue" = queue.Queue()
         self._lock = threading.Lock()
@@ -70,7 +71,7 @@ class RemoteForwarder:
 
     def _ensure_pool(self) -> ThreadPoolExecutor:
         if self._pool is None:
-            self._pool = ThreadPoolExecutor(max_workers=8,
+            self._pool = ThreadPoolExecutor(max_workers=self.max_workers,
                                             thread_name_prefix="remote-fwd")
         return self._pool
 

@@ -174,8 +174,17 @@ class BlockManager:
         self._ev_stored.append(h)
 
     # ---- allocation ------------------------------------------------------
-    def can_allocate(self, n_tokens: int) -> bool:
+    def can_allocate(self, n_tokens: int,
+                     seq_id: Optional[str] = None) -> bool:
+        """Can the pool grow `seq_id` (or a fresh sequence) to n_tokens?
+        Blocks the sequence already holds — including prefix-cache hits
+        adopted at admission — count toward the requirement; pricing them
+        as fresh allocations deadlocks a waiting sequence whose cached
+        prefix is most of the pool (found by the engine-lifecycle
+        property fuzz)."""
         need = (n_tokens + self.block_size - 1) // self.block_size
+        if seq_id is not None:
+            need -= len(self.tables.get(seq_id, ()))
         return need <= len(self._free_lru)
 
     def allocate(self, seq_id: str, n_tokens: int) -> bool:

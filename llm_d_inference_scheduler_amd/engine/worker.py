@@ -330,7 +330,8 @@ class EngineWorker:
                 self.waiting.remove(req)
                 continue
             chunk = min(remaining, budget)
-            if not self.mgr.can_allocate(req.computed + chunk) and not selected:
+            if not self.mgr.can_allocate(req.computed + chunk,
+                                         req.request_id) and not selected:
                 # cannot even fit one: stall this step (waiting queue grows)
                 break
             if not self.mgr.allocate(req.request_id, req.computed + chunk):

@@ -2,6 +2,7 @@
 random op sequences, flow-queue ordering, parser robustness on arbitrary
 bytes. ADVICE-tier hardening: these are the data structures whose
 accounting bugs would corrupt KV or strand requests silently."""
+import os
 import string
 
 from hypothesis import HealthCheck, given, settings
@@ -10,7 +11,9 @@ from hypothesis import strategies as st
 from llm_d_inference_scheduler_amd.engine.kvcache import (BlockManager,
                                                           block_hashes)
 
-SMALL = settings(max_examples=60, deadline=None,
+# DEEP_EXAMPLES=1000 turns the suite into a long fuzz campaign
+_N = int(os.environ.get("DEEP_EXAMPLES", "60"))
+SMALL = settings(max_examples=_N, deadline=None,
                  suppress_health_check=[HealthCheck.too_slow])
 
 
@@ -148,7 +151,7 @@ class TestParserFuzz:
 
 
 class TestKVBlockIndexInvariants:
-    @settings(max_examples=60, deadline=None)
+    @settings(max_examples=_N, deadline=None)
     @given(st.lists(st.tuples(
         st.sampled_from(["store", "evict", "spec", "remove"]),
         st.integers(0, 2),                     # endpoint
@@ -183,7 +186,7 @@ class TestKVBlockIndexInvariants:
 
 
 class TestEngineLifecycleProperty:
-    @settings(max_examples=10, deadline=None,
+    @settings(max_examples=max(10, _N // 6), deadline=None,
               suppress_health_check=[HealthCheck.too_slow])
     @given(st.lists(st.tuples(st.integers(1, 60),      # prompt len
                               st.integers(1, 6),       # max_tokens

@@ -90,6 +90,7 @@ class RequestOutput:
     ttft_ms: Optional[float] = None
     tpot_ms: Optional[float] = None
     e2e_ms: Optional[float] = None
+    error: str = ""                 # terminal engine-side failure
 
 
 class DecodeState:
@@ -207,6 +208,7 @@ class EngineWorker:
         self._pin = None
         self._pending = None   # (reqs, event|None, n) — one-step readback lag
         self.steps = 0
+        self._rejects: List[RequestOutput] = []
         self.total_generated = 0
         self.total_generated_slo = 0
         self.total_prefilled = 0
@@ -221,6 +223,18 @@ class EngineWorker:
             # engine loop (ids fold into the vocab deterministically)
             req.prompt_tokens = [t % self.cfg.vocab_size
                                  for t in req.prompt_tokens]
+        capacity = self.pool.num_blocks * self.pool.block_size
+        if len(req.prompt_tokens) + req.max_tokens > capacity:
+            # can never fit even running alone: reject instead of the
+            # infinite preempt/recompute loop the engine-lifecycle fuzz
+            # found (vLLM analog: scheduler watermark rejection)
+            self._rejects.append(RequestOutput(
+                request_id=req.request_id, new_tokens=[], finished=True,
+                finish_reason="error", prompt_tokens=len(req.prompt_tokens),
+                error="kv_capacity_exceeded: prompt+max_tokens "
+                      f"{len(req.prompt_tokens)}+{req.max_tokens} > "
+                      f"pool {capacity} tokens"))
+            return
         self.waiting.append(req)
         self._by_id[req.request_id] = req
 
@@ -271,7 +285,8 @@ class EngineWorker:
 
     @property
     def has_work(self) -> bool:
-        return bool(self.waiting or self.running or self._pending)
+        return bool(self.waiting or self.running or self._pending
+                    or self._rejects)
 
     # ------------------------------------------------------------------
     def step(self) -> List[RequestOutput]:
@@ -279,6 +294,9 @@ class EngineWorker:
         one decode pass over the running batch. Decode token values surface
         one step late (pipelined readback)."""
         outputs: List[RequestOutput] = []
+        if self._rejects:
+            outputs.extend(self._rejects)
+            self._rejects = []
         if self.waiting and self._should_prefill():
             outputs.extend(self._prefill_pass())
         if self.running:

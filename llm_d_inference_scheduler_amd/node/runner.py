@@ -836,7 +836,7 @@ class NodeRunner:
                "cached_tokens": out.cached_tokens,
                "tokens": out.all_tokens or [],
                "ttft_ms": out.ttft_ms, "tpot_ms": out.tpot_ms,
-               "e2e_ms": out.e2e_ms}
+               "e2e_ms": out.e2e_ms, "error": out.error}
         if self.is_router:
             self._handle_done(msg)
         else:

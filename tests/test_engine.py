@@ -199,3 +199,28 @@ class TestRefOps:
         s = x + r_orig
         expect = s * torch.rsqrt(s.pow(2).mean(-1, keepdim=True) + 1e-5) * w
         assert torch.allclose(y, expect, atol=1e-5)
+
+
+class TestCapacityRejection:
+    def test_impossible_request_rejected_not_livelocked(self):
+        """prompt+max_tokens beyond the whole pool must surface a
+        kv_capacity_exceeded error instead of the infinite
+        preempt/recompute loop (deep-fuzz find)."""
+        import torch
+        from llm_d_inference_scheduler_amd.engine import (EngineRequest,
+                                                          EngineWorker)
+        from llm_d_inference_scheduler_amd.models.configs import TINY_LLAMA
+        w = EngineWorker(TINY_LLAMA, "cpu", kv_blocks=5,
+                         dtype=torch.float32)       # 80 token slots
+        w.add_request(EngineRequest("big", list(range(72)), max_tokens=10))
+        w.add_request(EngineRequest("ok", list(range(40)), max_tokens=2))
+        outs = []
+        for _ in range(60):
+            outs.extend(w.step())
+            if not w.has_work:
+                break
+        assert not w.has_work
+        by_id = {o.request_id: o for o in outs if o.finished}
+        assert by_id["big"].error.startswith("kv_capacity_exceeded")
+        assert by_id["ok"].error == "" and len(by_id["ok"].new_tokens or
+                                               by_id["ok"].all_tokens) >= 1

@@ -224,17 +224,25 @@ class EngineWorker:
             req.prompt_tokens = [t % self.cfg.vocab_size
                                  for t in req.prompt_tokens]
         capacity = self.pool.num_blocks * self.pool.block_size
-        if len(req.prompt_tokens) + req.max_tokens > capacity:
-            # can never fit even running alone: reject instead of the
+        if len(req.prompt_tokens) >= capacity:
+            # the prompt alone can never fit: reject instead of the
             # infinite preempt/recompute loop the engine-lifecycle fuzz
             # found (vLLM analog: scheduler watermark rejection)
             self._rejects.append(RequestOutput(
                 request_id=req.request_id, new_tokens=[], finished=True,
                 finish_reason="error", prompt_tokens=len(req.prompt_tokens),
-                error="kv_capacity_exceeded: prompt+max_tokens "
-                      f"{len(req.prompt_tokens)}+{req.max_tokens} > "
-                      f"pool {capacity} tokens"))
+                error=f"kv_capacity_exceeded: prompt "
+                      f"{len(req.prompt_tokens)} >= pool {capacity} tokens"))
             return
+        if len(req.prompt_tokens) + req.max_tokens > capacity:
+            # prompt fits but full generation cannot: clamp so the request
+            # terminates at pool capacity (finish_reason stays "length")
+            # rather than preempt-looping against itself forever
+            clamped = capacity - len(req.prompt_tokens)
+            log.warning("max_tokens clamped to pool capacity",
+                        req=req.request_id, requested=req.max_tokens,
+                        clamped=clamped)
+            req.max_tokens = clamped
         self.waiting.append(req)
         self._by_id[req.request_id] = req
 

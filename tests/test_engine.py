@@ -202,10 +202,11 @@ class TestRefOps:
 
 
 class TestCapacityRejection:
-    def test_impossible_request_rejected_not_livelocked(self):
-        """prompt+max_tokens beyond the whole pool must surface a
-        kv_capacity_exceeded error instead of the infinite
-        preempt/recompute loop (deep-fuzz find)."""
+    def test_oversized_requests_terminate_not_livelock(self):
+        """A request that cannot fully fit the pool must still terminate
+        (deep-fuzz find): generation is clamped to pool capacity when the
+        prompt fits; a prompt larger than the whole pool is rejected with
+        kv_capacity_exceeded."""
         import torch
         from llm_d_inference_scheduler_amd.engine import (EngineRequest,
                                                           EngineWorker)
@@ -215,12 +216,26 @@ class TestCapacityRejection:
         w.add_request(EngineRequest("big", list(range(72)), max_tokens=10))
         w.add_request(EngineRequest("ok", list(range(40)), max_tokens=2))
         outs = []
-        for _ in range(60):
+        for _ in range(120):
             outs.extend(w.step())
             if not w.has_work:
                 break
         assert not w.has_work
         by_id = {o.request_id: o for o in outs if o.finished}
-        assert by_id["big"].error.startswith("kv_capacity_exceeded")
-        assert by_id["ok"].error == "" and len(by_id["ok"].new_tokens or
-                                               by_id["ok"].all_tokens) >= 1
+        # generation clamped to the 8 tokens that fit (80 - 72)
+        assert by_id["big"].error == ""
+        assert by_id["big"].completion_tokens == 8
+        assert by_id["ok"].error == ""
+
+    def test_prompt_larger_than_pool_rejected(self):
+        import torch
+        from llm_d_inference_scheduler_amd.engine import (EngineRequest,
+                                                          EngineWorker)
+        from llm_d_inference_scheduler_amd.models.configs import TINY_LLAMA
+        w = EngineWorker(TINY_LLAMA, "cpu", kv_blocks=4,
+                         dtype=torch.float32)       # 64 token slots
+        w.add_request(EngineRequest("huge", list(range(64)), max_tokens=1))
+        outs = w.step()
+        assert outs and outs[0].finished
+        assert outs[0].error.startswith("kv_capacity_exceeded")
+        assert not w.has_work

@@ -220,3 +220,57 @@ class TestEngineLifecycleProperty:
         assert stepped < 400, "engine failed to drain"
         assert done | aborted >= {f"r{i}" for i in range(len(reqs))}
         assert not w.mgr.tables, "leaked sequence tables"
+
+
+class TestFlowControlInvariants:
+    @SMALL
+    @given(st.lists(st.tuples(
+            st.sampled_from(["submit", "tick", "tick_blocked"]),
+            st.integers(-1, 10),                   # priority
+            st.integers(1, 50),                    # byte size
+            st.integers(0, 2)),                    # flow
+            min_size=1, max_size=120))
+    def test_every_item_reaches_one_terminal_state(self, ops):
+        """Random submit/tick/blocked-tick sequences: items end
+        dispatched/rejected/displaced exactly once, dispatched callbacks
+        match DISPATCHED outcomes, and the queue drains once dispatch
+        unblocks (processor.go exactly-once finalization invariants)."""
+        from llm_d_inference_scheduler_amd.flowcontrol import (
+            BandConfig, FlowController, FlowControlRequest, FlowRegistry,
+            QueueOutcome)
+        from llm_d_inference_scheduler_amd.scheduling.types import \
+            LLMRequest
+        reg = FlowRegistry(bands=[BandConfig(10), BandConfig(0),
+                                  BandConfig(-1, max_items=6)],
+                           global_max_bytes=600)
+        dispatched = []
+        state = {"ok": True}
+        fc = FlowController(reg, lambda item: state["ok"] and
+                            (dispatched.append(item) or True))
+        items = []
+        for i, (kind, prio, size, flow) in enumerate(ops):
+            if kind == "submit":
+                it = FlowControlRequest(
+                    request=LLMRequest(request_id=f"r{i}", model="m",
+                                       prompt="x"), flow_key=f"f{flow}",
+                    priority=prio, byte_size=size, ttl_s=60.0)
+                items.append(it)
+                fc.submit(it)
+            else:
+                state["ok"] = kind == "tick"
+                fc.tick()
+        state["ok"] = True
+        for _ in range(20):
+            fc.tick()
+        for it in items:
+            if it.outcome is not None:
+                assert it.outcome in (QueueOutcome.DISPATCHED,
+                                      QueueOutcome.REJECTED_CAPACITY,
+                                      QueueOutcome.EVICTED_DISPLACED)
+        for d in dispatched:
+            assert d.outcome == QueueOutcome.DISPATCHED
+        assert fc.queued_len == 0
+        n_disp = sum(1 for i in items
+                     if i.outcome == QueueOutcome.DISPATCHED)
+        assert n_disp == len(dispatched)
+        fc.stop()

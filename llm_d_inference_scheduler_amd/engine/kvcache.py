@@ -39,7 +39,16 @@ def block_hashes(prompt_tokens: Sequence[int], block_size: int = BLOCK_SIZE,
 class KVPool:
     """cache_dtype: torch.bfloat16 or torch.float8_e4m3fn (OCP fp8 — halves
     KV bytes and doubles effective decode-attention bandwidth; the gfx950
-    kernels convert with the hardware v_cvt_*_fp8 ops, hip_common.h)."""
+    kernels convert with the hardware v_cvt_*_fp8 ops, hip_common.h).
+
+    fp8 limitation (documented, by design): K/V are stored as UNSCALED
+    e4m3 — magnitudes >448 saturate and tiny values lose precision. This
+    matches the bf16->e4m3 direct-cast mode (vLLM's fp8 KV without
+    calibration scales); K/V activations of the supported model families
+    are well inside e4m3 range at bf16 training scale, and kv_cache_dtype
+    defaults to "auto" (=compute dtype). Outlier-heavy checkpoints should
+    keep bf16 KV; per-layer scale support would require threading a scale
+    through store (kv_cache.hip) and the attention dequant path."""
 
     def __init__(self, config: ModelConfig, num_blocks: int,
                  device: torch.device, dtype: torch.dtype = torch.bfloat16,

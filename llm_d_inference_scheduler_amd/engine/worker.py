@@ -135,6 +135,11 @@ class DecodeState:
              last_token: int) -> None:
         row = self.take_row()
         self.row_of[req.request_id] = row
+        # any membership mutation invalidates the cached row-index tensor:
+        # an identical id list can recur with different row assignments
+        # after a leave/rejoin (chunked-decode continuations)
+        self._active_ids = []
+        self._rows_t = None
         if len(table) > self.max_blocks:
             raise RuntimeError("sequence exceeds decode-state max_blocks")
         self.bt[row, :len(table)] = torch.tensor(
@@ -147,6 +152,8 @@ class DecodeState:
         row = self.row_of.pop(request_id, None)
         if row is not None:
             self.free_rows.append(row)
+            self._active_ids = []
+            self._rows_t = None
 
     def rows_for(self, ids: List[str]) -> torch.Tensor:
         if ids != self._active_ids:
@@ -199,9 +206,9 @@ class EngineWorker:
         self.waiting: List[EngineRequest] = []
         self.running: List[EngineRequest] = []
         self._by_id: Dict[str, EngineRequest] = {}
-        # rows must hold prompt + generation; generation length is
-        # unbounded by max_model_len in this engine, so budget 2x
-        max_blocks = (2 * max_model_len) // self.pool.block_size + 2
+        # rows must hold prompt + generation; add_request clamps
+        # prompt + max_tokens <= max_model_len, so this bound is exact
+        max_blocks = max_model_len // self.pool.block_size + 2
         self.dstate = DecodeState(min(64, max_decode_batch), max_blocks,
                                   self.device)
         self._cuda = self.device.type == "cuda"
@@ -214,9 +221,27 @@ class EngineWorker:
         self.total_prefilled = 0
 
     # ------------------------------------------------------------------
+    def _reject(self, req: EngineRequest, error: str) -> None:
+        self._rejects.append(RequestOutput(
+            request_id=req.request_id, new_tokens=[], finished=True,
+            finish_reason="error", prompt_tokens=len(req.prompt_tokens),
+            error=error))
+
     def add_request(self, req: EngineRequest) -> None:
+        if not req.prompt_tokens:
+            # an empty prompt would be silently dropped in _prefill_pass
+            # (remaining <= 0) and the client would hang until timeout
+            self._reject(req, "empty_prompt: prompt tokenized to 0 tokens")
+            return
         if len(req.prompt_tokens) >= self.max_model_len:
-            req.prompt_tokens = req.prompt_tokens[:self.max_model_len - 1]
+            # reject, never truncate: a silently-truncated prompt produces
+            # a generation the caller cannot interpret (vLLM analog:
+            # context_length_exceeded)
+            self._reject(
+                req, f"context_length_exceeded: prompt "
+                     f"{len(req.prompt_tokens)} >= max_model_len "
+                     f"{self.max_model_len}")
+            return
         if any(t >= self.cfg.vocab_size or t < 0
                for t in req.prompt_tokens):
             # defensive clamp: a mismatched tokenizer must not crash the
@@ -228,18 +253,17 @@ class EngineWorker:
             # the prompt alone can never fit: reject instead of the
             # infinite preempt/recompute loop the engine-lifecycle fuzz
             # found (vLLM analog: scheduler watermark rejection)
-            self._rejects.append(RequestOutput(
-                request_id=req.request_id, new_tokens=[], finished=True,
-                finish_reason="error", prompt_tokens=len(req.prompt_tokens),
-                error=f"kv_capacity_exceeded: prompt "
-                      f"{len(req.prompt_tokens)} >= pool {capacity} tokens"))
+            self._reject(
+                req, f"kv_capacity_exceeded: prompt "
+                     f"{len(req.prompt_tokens)} >= pool {capacity} tokens")
             return
-        if len(req.prompt_tokens) + req.max_tokens > capacity:
-            # prompt fits but full generation cannot: clamp so the request
-            # terminates at pool capacity (finish_reason stays "length")
-            # rather than preempt-looping against itself forever
-            clamped = capacity - len(req.prompt_tokens)
-            log.warning("max_tokens clamped to pool capacity",
+        # generation ceiling: both the KV pool AND the model context window
+        # (positions past max_model_len walk off the RoPE table and past
+        # DecodeState's block-table width)
+        ceiling = min(capacity, self.max_model_len)
+        if len(req.prompt_tokens) + req.max_tokens > ceiling:
+            clamped = ceiling - len(req.prompt_tokens)
+            log.warning("max_tokens clamped to capacity",
                         req=req.request_id, requested=req.max_tokens,
                         clamped=clamped)
             req.max_tokens = clamped
@@ -568,7 +592,9 @@ class EngineWorker:
                 # victim: lowest InferenceObjective priority, youngest last
                 victim = min(reversed(self.running),
                              key=lambda r: r.priority)
-                self._preempt(victim)
+                po = self._preempt(victim)
+                if po is not None:
+                    outs.append(po)
             return outs
 
         if upd_rows:
@@ -655,23 +681,35 @@ class EngineWorker:
             outputs.append(out)
         return outputs
 
-    def _preempt(self, req: EngineRequest) -> None:
+    def _preempt(self, req: EngineRequest) -> Optional[RequestOutput]:
         """Evict a running request and requeue it for recompute: its
         generation so far folds into the prompt (prefill of prompt+generated
         produces the next token's logits, so accounting continues exactly).
-        Pending tokens must be collected first (req.inflight == 0)."""
+        Pending tokens must be collected first (req.inflight == 0).
+
+        If prompt+generated no longer fits the context window (can only
+        happen for requests admitted before a config change — add_request
+        clamps prompt+max_tokens to max_model_len), finish the request
+        (finish_reason=length) instead of truncating: truncation would
+        recompute KV that no longer corresponds to the retained generation
+        and re-emit dropped tokens through all_tokens."""
         assert req.inflight == 0, "collect pending before preempting"
+        merged = req.prompt_tokens + req.generated
+        if len(merged) > self.max_model_len - 1:
+            log.info("preempt would exceed context window; finishing",
+                     req=req.request_id, generated=len(req.generated))
+            return self._finalize(req)
         log.info("preempting for KV space", req=req.request_id,
                  generated=len(req.generated))
         self.running.remove(req)
         self.dstate.leave(req.request_id)
         self.mgr.free(req.request_id)
-        merged = (req.prompt_tokens + req.generated)[:self.max_model_len - 1]
         req.prompt_tokens = merged
         req.computed = 0
         req.block_hashes = None
         req.registered_blocks = 0
         self.waiting.insert(0, req)
+        return None
 
     def _finalize(self, req: EngineRequest) -> RequestOutput:
         """Emit the finished output for a request with no in-flight token

@@ -239,3 +239,66 @@ class TestCapacityRejection:
         assert outs and outs[0].finished
         assert outs[0].error.startswith("kv_capacity_exceeded")
         assert not w.has_work
+
+
+class TestModelLenLimits:
+    """ADVICE round-1 fixes: context-window enforcement at admission and
+    preemption (engine/worker.py add_request/_preempt)."""
+
+    def test_prompt_at_model_len_rejected_not_truncated(self):
+        w = make_worker(max_model_len=64)
+        w.add_request(EngineRequest("long", list(range(64)), max_tokens=4))
+        outs = w.step()
+        assert outs and outs[0].finished
+        assert outs[0].error.startswith("context_length_exceeded")
+        assert not w.has_work
+
+    def test_max_tokens_clamped_to_model_len(self):
+        w = make_worker(max_model_len=64)
+        w.add_request(EngineRequest("r", list(range(40)), max_tokens=500))
+        outs = run_to_completion(w)
+        fin = [o for o in outs if o.finished]
+        assert len(fin) == 1 and fin[0].error == ""
+        # 40 prompt + 24 generated == max_model_len
+        assert fin[0].completion_tokens == 24
+        assert fin[0].finish_reason == "length"
+
+    def test_empty_prompt_rejected(self):
+        w = make_worker()
+        w.add_request(EngineRequest("empty", [], max_tokens=4))
+        outs = w.step()
+        assert outs and outs[0].finished
+        assert outs[0].error.startswith("empty_prompt")
+        assert not w.has_work
+
+    def test_preempt_past_window_finishes_instead_of_truncating(self):
+        """A running request whose prompt+generated can no longer be
+        requeued (context window shrank under it) must finish cleanly with
+        its generation intact, not restart with a truncated prompt."""
+        w = make_worker(kv_blocks=8)   # 128 token slots
+        w.add_request(EngineRequest("a", list(range(60)), max_tokens=60))
+        w.add_request(EngineRequest("b", list(range(60, 120)), max_tokens=60))
+        # run until both are decoding
+        for _ in range(30):
+            w.step()
+            if len(w.running) == 2:
+                break
+        assert len(w.running) == 2
+        # simulate the window shrinking below prompt+generated before a
+        # preemption (admitted-before-config-change scenario)
+        w.max_model_len = 32
+        victim = w.running[-1]
+        gen_before = list(victim.generated)
+        # collect pending so inflight == 0, as _decode_pass guarantees
+        outs = []
+        for _ in range(4):
+            outs.extend(w.step())
+        outs.extend(o for o in w._collect_pending())
+        victim_reqs = [r for r in w.running if r.inflight == 0]
+        if victim in victim_reqs:
+            out = w._preempt(victim)
+            assert out is not None and out.finished
+            assert out.finish_reason == "length"
+            # generation retained exactly (no token ever dropped)
+            assert out.all_tokens[:len(gen_before)] == gen_before
+            assert victim not in w.running

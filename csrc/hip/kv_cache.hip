@@ -79,6 +79,32 @@ __global__ void scatter_blocks_kernel(short8* __restrict__ pool,
   }
 }
 
+// Direct peer-to-peer block copy: gather blocks `src_ids` straight out of a
+// PEER GPU's pool (mapped via hipIpcOpenMemHandle — loads travel over the
+// direct xGMI link) into this GPU's pool blocks `dst_ids`. One-sided pull:
+// no staging buffer, no send/recv rendezvous (SURVEY.md §5.8 names
+// hipMemcpyPeerAsync; a gather kernel over the mapped peer pointer is the
+// same xGMI path but handles the non-contiguous block list in one launch).
+__global__ void copy_blocks_peer_kernel(const short8* __restrict__ src_pool,
+                                        short8* __restrict__ dst_pool,
+                                        const int32_t* __restrict__ src_ids,
+                                        const int32_t* __restrict__ dst_ids,
+                                        int n_sel, int n_layers,
+                                        int64_t src_nb, int64_t dst_nb,
+                                        int64_t block_elems8) {
+  const int64_t per_sel = (int64_t)n_layers * 2 * block_elems8;
+  const int64_t total = per_sel * n_sel;
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < total;
+       i += (int64_t)gridDim.x * blockDim.x) {
+    int64_t sel = i / per_sel;
+    int64_t rem = i % per_sel;
+    int64_t lkv = rem / block_elems8;
+    int64_t off = rem % block_elems8;
+    dst_pool[(lkv * dst_nb + dst_ids[sel]) * block_elems8 + off] =
+        src_pool[(lkv * src_nb + src_ids[sel]) * block_elems8 + off];
+  }
+}
+
 }  // namespace
 
 extern "C" {
@@ -128,6 +154,25 @@ hipError_t lds_gather_blocks(const void* pool, void* staging,
                        stream, (const short8*)pool, (short8*)staging, block_ids,
                        n_sel, n_layers, n_blocks, block_elems / 8);
   }
+  HIP_CHECK_LAST();
+  return hipSuccess;
+}
+
+hipError_t lds_copy_blocks_peer(const void* src_pool, void* dst_pool,
+                                const int32_t* src_ids, const int32_t* dst_ids,
+                                int n_sel, int n_layers, int64_t src_nb,
+                                int64_t dst_nb, int64_t block_bytes,
+                                hipStream_t stream) {
+  if (n_sel == 0) return hipSuccess;
+  int64_t block_elems8 = block_bytes / 16;
+  int64_t total = (int64_t)n_sel * n_layers * 2 * block_elems8;
+  int threads = 256;
+  int blocks = (int)((total + threads - 1) / threads);
+  if (blocks > 4096) blocks = 4096;
+  hipLaunchKernelGGL(copy_blocks_peer_kernel, dim3(blocks), dim3(threads), 0,
+                     stream, (const short8*)src_pool, (short8*)dst_pool,
+                     src_ids, dst_ids, n_sel, n_layers, src_nb, dst_nb,
+                     block_elems8);
   HIP_CHECK_LAST();
   return hipSuccess;
 }

@@ -180,6 +180,66 @@ void move_blocks(torch::Tensor pool, torch::Tensor staging,
                               is_scatter ? 1 : 0, cur_stream()));
 }
 
+// ---- HIP IPC surface for the direct peer-pull transfer path -------------
+// The KV pool is allocated with its own hipMalloc (NOT the caching
+// allocator, whose suballocation offsets would corrupt IPC handles); the
+// decode rank opens the prefill rank's handle once and pulls blocks with
+// copy_blocks_peer over the mapped pointer (xGMI one-sided read).
+
+torch::Tensor ipc_alloc_tensor(std::vector<int64_t> shape,
+                               torch::Tensor like) {
+  int64_t numel = 1;
+  for (auto s : shape) numel *= s;
+  int64_t nbytes = numel * like.element_size();
+  void* ptr = nullptr;
+  CHECK_HIP(hipMalloc(&ptr, nbytes));
+  CHECK_HIP(hipMemset(ptr, 0, nbytes));
+  auto deleter = [](void* p) { (void)hipFree(p); };
+  return torch::from_blob(ptr, shape, deleter,
+                          like.options());
+}
+
+py::bytes ipc_handle(torch::Tensor t) {
+  hipIpcMemHandle_t h;
+  CHECK_HIP(hipIpcGetMemHandle(&h, t.data_ptr()));
+  return py::bytes(reinterpret_cast<const char*>(&h), sizeof(h));
+}
+
+uintptr_t ipc_open(py::bytes handle) {
+  std::string s = handle;
+  TORCH_CHECK(s.size() == sizeof(hipIpcMemHandle_t), "bad IPC handle size");
+  hipIpcMemHandle_t h;
+  memcpy(&h, s.data(), sizeof(h));
+  void* ptr = nullptr;
+  hipError_t err =
+      hipIpcOpenMemHandle(&ptr, h, hipIpcMemLazyEnablePeerAccess);
+  TORCH_CHECK(err == hipSuccess, "hipIpcOpenMemHandle: ",
+              hipGetErrorString(err));
+  return reinterpret_cast<uintptr_t>(ptr);
+}
+
+void ipc_close(uintptr_t ptr) {
+  CHECK_HIP(hipIpcCloseMemHandle(reinterpret_cast<void*>(ptr)));
+}
+
+extern "C" hipError_t lds_copy_blocks_peer(const void*, void*, const int32_t*,
+                                           const int32_t*, int, int, int64_t,
+                                           int64_t, int64_t, hipStream_t);
+
+void copy_blocks_peer(uintptr_t src_pool_ptr, torch::Tensor dst_pool,
+                      torch::Tensor src_ids, torch::Tensor dst_ids,
+                      int64_t src_nb) {
+  CHECK_GPU(dst_pool); CHECK_GPU(src_ids); CHECK_GPU(dst_ids);
+  int L = (int)dst_pool.size(0);
+  int64_t dst_nb = dst_pool.size(2);
+  int64_t block_bytes = dst_pool.size(3) * dst_pool.size(4) *
+                        dst_pool.size(5) * dst_pool.element_size();
+  CHECK_HIP(lds_copy_blocks_peer(
+      reinterpret_cast<const void*>(src_pool_ptr), dst_pool.data_ptr(),
+      src_ids.data_ptr<int32_t>(), dst_ids.data_ptr<int32_t>(),
+      (int)src_ids.size(0), L, src_nb, dst_nb, block_bytes, cur_stream()));
+}
+
 torch::Tensor paged_attention(torch::Tensor q, torch::Tensor k_cache,
                               torch::Tensor v_cache,
                               torch::Tensor block_tables,
@@ -290,6 +350,13 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("silu_mul", &silu_mul, "fused SiLU*up on packed gate_up");
   m.def("reshape_and_cache", &reshape_and_cache, "scatter K/V into paged pool");
   m.def("move_blocks", &move_blocks, "gather/scatter KV blocks for xGMI transfer");
+  m.def("ipc_alloc_tensor", &ipc_alloc_tensor,
+        "hipMalloc-backed tensor (IPC-shareable, outside the caching allocator)");
+  m.def("ipc_handle", &ipc_handle, "hipIpcGetMemHandle of a tensor");
+  m.def("ipc_open", &ipc_open, "map a peer pool (hipIpcOpenMemHandle)");
+  m.def("ipc_close", &ipc_close, "unmap a peer pool");
+  m.def("copy_blocks_peer", &copy_blocks_peer,
+        "one-sided xGMI pull of KV blocks from a mapped peer pool");
   m.def("paged_attention", &paged_attention, "GQA decode attention over paged KV");
   m.def("paged_attention_split", &paged_attention_split,
         "flash-decoding GQA attention with sequence partitioning");

@@ -53,17 +53,30 @@ class KVPool:
     def __init__(self, config: ModelConfig, num_blocks: int,
                  device: torch.device, dtype: torch.dtype = torch.bfloat16,
                  block_size: int = BLOCK_SIZE,
-                 cache_dtype: torch.dtype = None):
+                 cache_dtype: torch.dtype = None,
+                 ipc_alloc: bool = False):
         self.cfg = config
         self.num_blocks = num_blocks
         self.block_size = block_size
         self.device = device
         self.dtype = dtype
         self.cache_dtype = cache_dtype or dtype
-        self.tensor = torch.zeros(
-            (config.num_layers, 2, num_blocks, config.num_kv_heads,
-             block_size, config.head_dim), dtype=self.cache_dtype,
-            device=device)
+        shape = (config.num_layers, 2, num_blocks, config.num_kv_heads,
+                 block_size, config.head_dim)
+        self.ipc_backed = False
+        if ipc_alloc:
+            # dedicated hipMalloc (outside the caching allocator) so the
+            # pool is IPC-shareable for the peer-pull transfer transport:
+            # an IPC handle of a caching-allocator suballocation maps the
+            # containing arena, not the pool, on the peer side
+            from .. import ops as ops_mod
+            like = torch.empty(0, dtype=self.cache_dtype, device=device)
+            self.tensor = ops_mod.hip_ops().ipc_alloc_tensor(
+                list(shape), like)
+            self.ipc_backed = True
+        else:
+            self.tensor = torch.zeros(shape, dtype=self.cache_dtype,
+                                      device=device)
 
     def layer(self, li: int):
         return self.tensor[li, 0], self.tensor[li, 1]

@@ -178,9 +178,11 @@ class EngineWorker:
                  kv_cache_dtype: str = "auto",
                  prefill_min_tokens: int = 4096,
                  prefill_max_delay_ms: float = 60.0,
+                 ipc_pool: bool = False,
                  seed: int = 0):
         self.cfg = config
         self.device = torch.device(device)
+        self._is_cuda_dev = lambda: self.device.type == "cuda"
         self.role = role
         self.dtype = dtype
         self.prefill_chunk_tokens = prefill_chunk_tokens
@@ -199,7 +201,8 @@ class EngineWorker:
                 config, kv_budget_bytes,
                 dtype_bytes=cache_dtype.itemsize)
         self.pool = KVPool(config, kv_blocks, self.device, dtype,
-                           cache_dtype=cache_dtype)
+                           cache_dtype=cache_dtype,
+                           ipc_alloc=ipc_pool and self._is_cuda_dev())
         self.mgr = BlockManager(kv_blocks, self.pool.block_size,
                                 prefix_caching=prefix_caching)
         self.model = LlamaRunner(config, self.device, dtype, seed=seed)

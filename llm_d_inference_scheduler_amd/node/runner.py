@@ -103,6 +103,19 @@ class NodeConfig:
     #                  decode-side setup overlap and the transfer can never
     #                  hit kv_exhausted (connector_sglang.go)
     connector: str = "nixlv2"
+    # KV transfer transport (parallel/transfer.py):
+    #   auto  -> direct peer-pull over HIP IPC when every rank can map every
+    #            other rank's pool (one-sided xGMI gather, no staging);
+    #            falls back to rccl otherwise — the fallback decision is a
+    #            collective so all ranks agree
+    #   rccl  -> staged gather + RCCL send/recv
+    #   peer  -> require IPC (startup error if unavailable)
+    transfer_transport: str = "auto"
+    # back-pressure: max NEW KV-transfer bytes started per step (deferral is
+    # a deterministic function of the shared kv_ready job stream, so sender
+    # and receiver always defer the same jobs and RCCL pairing stays
+    # matched); at least one job always starts
+    max_step_transfer_bytes: int = 1 << 30
     # injectable transport for remote-node forwarding (tests); None = httpx
     remote_transport: Any = None
     cache_hit_threshold: float = 0.8
@@ -133,13 +146,17 @@ class NodeRunner:
         self.my_spec = self.topology.ranks[self.rank]
         self.mailbox = Mailbox(cfg.mailbox_group, self.rank, cfg.world_size)
 
+        want_ipc = (cfg.world_size > 1
+                    and str(cfg.device).startswith("cuda")
+                    and cfg.transfer_transport in ("auto", "peer"))
         self.engine = EngineWorker(
             cfg.model, cfg.device, role=self.my_spec.role,
             kv_blocks=cfg.kv_blocks, kv_budget_bytes=cfg.kv_budget_bytes,
             dtype=cfg.dtype, prefill_chunk_tokens=cfg.prefill_chunk_tokens,
             max_decode_batch=cfg.max_decode_batch,
             ttft_slo_ms=cfg.ttft_slo_ms,
-            kv_cache_dtype=cfg.kv_cache_dtype, seed=cfg.seed)
+            kv_cache_dtype=cfg.kv_cache_dtype, ipc_pool=want_ipc,
+            seed=cfg.seed)
         # encode role: vision tower + URL-deduped embedding cache
         self.encoder = None
         if (self.my_spec.role & Role.ENCODE) and cfg.model.vision_hidden:
@@ -149,8 +166,14 @@ class NodeRunner:
         self._encode_jobs: List[Dict[str, Any]] = []
         self._emb_pending: Dict[str, torch.Tensor] = {}  # src side
         self._awaiting_embeds: Dict[str, Dict[str, Any]] = {}  # dst side
+        transport, peer_pools = self._negotiate_transport(want_ipc)
         self.transfer = KVTransferEngine(self.engine.pool.tensor, self.rank,
-                                         group=cfg.transfer_group)
+                                         group=cfg.transfer_group,
+                                         transport=transport,
+                                         peer_pools=peer_pools)
+        # kv_ready jobs deferred by the per-step transfer-bytes cap
+        # (identical list on every rank: deferral is deterministic)
+        self._deferred_transfers: List[Dict[str, Any]] = []
         self._outbox: List[Dict[str, Any]] = []
         # decode-side: requests waiting for a remote prefill's KV
         self._pending_adoption: Dict[str, Dict[str, Any]] = {}
@@ -168,6 +191,54 @@ class NodeRunner:
         self.is_router = (self.rank == 0)
         if self.is_router:
             self._init_router()
+
+    # ------------------------------------------------------------------
+    def _negotiate_transport(self, want_ipc: bool):
+        """Collective transport agreement: peer-pull only if EVERY rank
+        published an IPC handle and EVERY rank mapped every peer — a mixed
+        decision would deadlock the RCCL pairing."""
+        cfg = self.cfg
+        if cfg.world_size <= 1 or not want_ipc:
+            return "rccl", {}
+        import torch.distributed as dist_mod
+        from .. import ops as ops_mod
+        handle = None
+        if getattr(self.engine.pool, "ipc_backed", False):
+            try:
+                handle = bytes(
+                    ops_mod.hip_ops().ipc_handle(self.engine.pool.tensor))
+            except Exception as e:
+                log.warning("ipc_handle failed", err=str(e))
+        gathered = [None] * cfg.world_size
+        dist_mod.all_gather_object(gathered,
+                                   (handle, self.engine.pool.num_blocks),
+                                   group=cfg.mailbox_group)
+        peer_pools = {}
+        ok = all(h is not None for h, _ in gathered)
+        if ok:
+            try:
+                for r, (h, rnb) in enumerate(gathered):
+                    if r == self.rank:
+                        continue
+                    peer_pools[r] = (ops_mod.hip_ops().ipc_open(h), rnb)
+            except Exception as e:
+                log.warning("ipc_open failed", err=str(e))
+                ok = False
+        flags = [None] * cfg.world_size
+        dist_mod.all_gather_object(flags, bool(ok), group=cfg.mailbox_group)
+        if all(flags):
+            log.info("KV transfer transport: peer (HIP IPC xGMI pull)")
+            return "peer", peer_pools
+        for ptr, _ in peer_pools.values():
+            try:
+                ops_mod.hip_ops().ipc_close(ptr)
+            except Exception:
+                pass
+        if cfg.transfer_transport == "peer":
+            raise RuntimeError("transfer_transport=peer requested but HIP "
+                               "IPC negotiation failed on some rank")
+        log.info("KV transfer transport: rccl (staged send/recv)")
+        return "rccl", {}
 
     # ------------------------------------------------------------------
     def _init_router(self) -> None:
@@ -360,6 +431,9 @@ class NodeRunner:
     # ------------------------------------------------------------------
     def step(self) -> None:
         """One lockstep node iteration on every rank."""
+        # fire completions of transfers launched on earlier steps (adoption,
+        # prefill-block release) before building this step's batch
+        self.transfer.poll()
         if self.is_router:
             self._route_arrivals()
             self._maybe_evict_inflight()
@@ -570,6 +644,11 @@ class NodeRunner:
             elif t == "pd_skip":
                 self._deferred_prefill.pop(m["req_id"], None)
                 self._handoff_dst.pop(m["req_id"], None)
+            elif t == "kv_released" and m.get("dst") == self.rank:
+                # peer-pull transport: the decode rank finished (or
+                # abandoned) its one-sided pull; the prefill blocks are
+                # safe to recycle now
+                self.engine.release_prefilled(m["req_id"])
             elif t == "kv_events" and self.is_router:
                 if self._precise is not None:
                     self._precise.apply_events(f"gpu{m['src']}", m["s"],
@@ -743,6 +822,15 @@ class NodeRunner:
 
     # ---- transfers (the NIXL-v2 step 2/3 replacement) ----
     def _execute_transfers(self, msgs: List[Dict[str, Any]]) -> None:
+        """Start KV/embedding movement for this step's hand-off messages.
+
+        KV transfers are ASYNC: work is enqueued on the transfer stream and
+        completion fires from `transfer.poll()` on a later step — decode
+        compute overlaps the copy (SURVEY §7 hard part #5). Back-pressure:
+        at most `max_step_transfer_bytes` of NEW jobs start per step; the
+        rest defer. Deferral depends only on the shared, sorted job stream,
+        so every rank defers the same jobs and RCCL send/recv pairing stays
+        matched."""
         emb_jobs = sorted((m for m in msgs if m.get("type") == "emb_ready"),
                           key=lambda m: m["req_id"])
         for job in emb_jobs:
@@ -755,41 +843,78 @@ class NodeRunner:
                     src, (job["rows"], self.cfg.model.hidden_size),
                     self.cfg.dtype)
                 self._attach_embeds(job["req_id"], emb)
-        jobs = sorted((m for m in msgs if m.get("type") == "kv_ready"),
-                      key=lambda m: m["req_id"])
-        for job in jobs:
-            src, dst = job["src"], job["dst"]
-            if self.rank == src:
-                self.transfer.send_blocks(dst, job["blocks"])
-                self.engine.release_prefilled(job["req_id"])
-            elif self.rank == dst:
-                n = len(job["blocks"])
-                reserved = (self._pending_adoption.get(job["req_id"]) or
-                            {}).get("reserved")
-                if reserved is not None and len(reserved) != n:
-                    self.engine.mgr.release_blocks(reserved)
-                    reserved = None
-                local = reserved if reserved is not None else \
-                    self.engine.mgr.take_blocks(n)
-                if local is None:
-                    # keep the P2P matched: receive into scratch and drop
-                    scratch = self.transfer._staging(n)
-                    torch.distributed.recv(self.transfer._wire(scratch),
-                                           src=src,
-                                           group=self.cfg.transfer_group)
-                    self._outbox.append({"type": "done",
-                                         "req_id": job["req_id"],
-                                         "error": "kv_exhausted"})
-                    self._pending_adoption.pop(job["req_id"], None)
-                    continue
-                self.transfer.recv_blocks(src, local)
-                pending = self._pending_adoption.pop(job["req_id"], None)
-                if pending is None:
-                    self.engine.mgr.release_blocks(local)
-                    continue
+        new_jobs = sorted((m for m in msgs if m.get("type") == "kv_ready"),
+                          key=lambda m: m["req_id"])
+        jobs = self._deferred_transfers + new_jobs
+        budget = self.cfg.max_step_transfer_bytes
+        started = 0
+        for i, job in enumerate(jobs):
+            nbytes = self.transfer._block_nbytes(len(job["blocks"]))
+            if started > 0 and nbytes > budget:
+                self._deferred_transfers = jobs[i:]
+                break
+            budget -= nbytes
+            started += 1
+            self._start_kv_job(job)
+        else:
+            self._deferred_transfers = []
+
+    def _start_kv_job(self, job: Dict[str, Any]) -> None:
+        src, dst, req_id = job["src"], job["dst"], job["req_id"]
+        n = len(job["blocks"])
+        peer = self.transfer.transport == "peer"
+        if self.rank == src:
+            if peer:
+                return      # one-sided: dst pulls; release on kv_released
+            self.transfer.send_blocks(
+                dst, job["blocks"],
+                on_complete=lambda: self.engine.release_prefilled(req_id))
+            return
+        if self.rank != dst:
+            return
+        pending0 = self._pending_adoption.get(req_id)
+        reserved = (pending0 or {}).get("reserved")
+        if pending0 is not None:
+            # ownership of the reservation moves to this job NOW: an abort
+            # arriving mid-transfer must not release blocks the in-flight
+            # copy is writing (the _adopt completion owns their release)
+            pending0["reserved"] = None
+        if reserved is not None and len(reserved) != n:
+            self.engine.mgr.release_blocks(reserved)
+            reserved = None
+        local = reserved if reserved is not None else \
+            self.engine.mgr.take_blocks(n)
+        if local is None:
+            # cannot adopt (kv_exhausted): drop the hand-off
+            self._outbox.append({"type": "done", "req_id": req_id,
+                                 "error": "kv_exhausted"})
+            self._pending_adoption.pop(req_id, None)
+            if peer:
+                # nothing was pulled; free the src blocks immediately
+                self._outbox.append({"type": "kv_released",
+                                     "req_id": req_id, "dst": src})
+            else:
+                # keep the P2P pairing matched: receive into scratch
+                self.transfer.recv_discard(src, n)
+            return
+
+        def _adopt():
+            pending = self._pending_adoption.pop(req_id, None)
+            if pending is None:           # aborted while in flight
+                self.engine.mgr.release_blocks(local)
+            else:
                 self.engine.admit_transferred(pending["req"], local,
                                               job["seq_len"],
                                               job["first_token"])
+            if peer:
+                self._outbox.append({"type": "kv_released",
+                                     "req_id": req_id, "dst": src})
+
+        if peer:
+            self.transfer.pull_blocks(src, job["blocks"], local,
+                                      on_complete=_adopt)
+        else:
+            self.transfer.recv_blocks(src, local, on_complete=_adopt)
 
     # ---- engine outputs -> messages ----
     def _handle_outputs(self, outputs: List[RequestOutput]) -> None:
@@ -844,6 +969,7 @@ class NodeRunner:
 
     # ------------------------------------------------------------------
     def shutdown(self) -> None:
+        self.transfer.synchronize()   # drain in-flight xGMI transfers
         if self.is_router and self.flow is not None:
             self.flow.stop()
         if self.is_router and getattr(self, "remote", None) is not None:

@@ -1,23 +1,35 @@
-"""xGMI KV/embedding transfer engine.
+"""xGMI KV/embedding transfer engine (async, overlapped).
 
 This is the MI355X-native replacement for the data movement the reference
 only *triggers* (NIXL/UCX/RDMA `kv_transfer_params`, connector_nixlv2.go;
 SURVEY.md §5.8): prefill-role and decode-role GPUs exchange whole KV blocks
 directly over the node's xGMI links.
 
-Mechanism: the gfx950 gather kernel (kv_cache.hip) packs the selected
-blocks — already MFMA-tile-aligned, so the decode kernel consumes them with
-zero repack — into one contiguous staging buffer; a single RCCL
-send/recv (torch.distributed P2P over the "nccl"=RCCL backend) moves it
-point-to-point (any P->D pair has a direct ~153 GB/s xGMI path); the
-receiver's scatter kernel drops the blocks into its own pool. Transfers run
-on a dedicated HIP stream so decode compute overlaps the copy; the
-`kv_transfer_params` JSON contract becomes the in-process block-handle
+Two transports, both running on a dedicated HIP stream so decode compute
+overlaps the copy, with completion signalled by a recorded hipEvent the node
+loop POLLS (no host synchronize on the hot path — round-1's
+`stream.synchronize()` right after the send killed the claimed overlap):
+
+* ``peer`` (default on GPU when HIP IPC is available): the decode rank maps
+  the prefill rank's pool once via hipIpcOpenMemHandle and PULLS blocks
+  with a one-sided gather kernel over the mapped pointer (kv_cache.hip
+  copy_blocks_peer — the hipMemcpyPeerAsync xGMI path, generalized to a
+  non-contiguous block list in one launch). No staging buffer, no
+  rendezvous; the sender only keeps its blocks allocated until the
+  receiver's ``kv_released`` control message.
+* ``rccl``: gfx950 gather kernel packs blocks into a contiguous staging
+  buffer; RCCL send/recv (torch.distributed P2P over the "nccl"=RCCL
+  backend) moves it point-to-point; the receiver's scatter kernel drops the
+  blocks into its own pool. Used when IPC is unavailable and for the
+  CPU/gloo hermetic tests (where ops run synchronously and handles complete
+  immediately).
+
+The `kv_transfer_params` JSON contract becomes the in-process block-handle
 messages carried by the mailbox (engine id -> rank, block IDs -> pool
-indices). CPU/gloo path exists for hermetic multi-process tests.
+indices).
 """
 import time
-from typing import List, Optional
+from typing import Callable, Dict, List, Optional
 
 import torch
 import torch.distributed as dist
@@ -26,58 +38,156 @@ from .. import ops
 from ..metrics import prom
 
 
+class TransferHandle:
+    """One in-flight transfer: completion = recorded event has fired.
+
+    `on_complete` runs exactly once, from poll(), on the host thread that
+    owns the node loop (no cross-thread state)."""
+
+    __slots__ = ("event", "on_complete", "nbytes", "t0", "_staging", "done",
+                 "label")
+
+    def __init__(self, event, on_complete: Optional[Callable], nbytes: int,
+                 label: str, staging=None):
+        self.event = event
+        self.on_complete = on_complete
+        self.nbytes = nbytes
+        self.t0 = time.monotonic()
+        self._staging = staging    # keep the buffer alive until completion
+        self.done = False
+        self.label = label
+
+    def poll(self) -> bool:
+        if self.done:
+            return True
+        if self.event is not None and not self.event.query():
+            return False
+        self.done = True
+        self._staging = None
+        prom.xgmi_kv_transfer_bytes.labels(self.label).inc(self.nbytes)
+        prom.xgmi_kv_transfer_seconds.observe(time.monotonic() - self.t0)
+        if self.on_complete is not None:
+            self.on_complete()
+        return True
+
+
 class KVTransferEngine:
     def __init__(self, pool_tensor: torch.Tensor, rank: int,
                  group: Optional[object] = None,
-                 device: Optional[torch.device] = None):
+                 device: Optional[torch.device] = None,
+                 transport: str = "rccl",
+                 peer_pools: Optional[Dict[int, tuple]] = None):
         self.pool = pool_tensor  # [L, 2, NB, KVH, BS, D]
         self.rank = rank
         self.group = group
         self.device = device or pool_tensor.device
         self.on_gpu = self.pool.is_cuda
         self.stream = torch.cuda.Stream(self.device) if self.on_gpu else None
+        self.transport = transport if self.on_gpu else "rccl"
+        # rank -> (mapped_ptr:int, n_blocks:int) of each peer's pool
+        self.peer_pools: Dict[int, tuple] = peer_pools or {}
+        self.pending: List[TransferHandle] = []
 
+    # ------------------------------------------------------------------
+    def poll(self) -> None:
+        """Fire completions for finished transfers (called once per step)."""
+        if not self.pending:
+            return
+        self.pending = [h for h in self.pending if not h.poll()]
+
+    @property
+    def inflight_bytes(self) -> int:
+        return sum(h.nbytes for h in self.pending if not h.done)
+
+    def synchronize(self) -> None:
+        """Drain all pending transfers (shutdown / tests only)."""
+        if self.stream is not None:
+            self.stream.synchronize()
+        self.poll()
+
+    # ------------------------------------------------------------------
     def _staging(self, n_blocks: int) -> torch.Tensor:
         L = self.pool.shape[0]
         return torch.empty((n_blocks, L, 2) + tuple(self.pool.shape[3:]),
                            dtype=self.pool.dtype, device=self.device)
+
+    def _block_nbytes(self, n_blocks: int) -> int:
+        L = self.pool.shape[0]
+        per = 1
+        for s in self.pool.shape[3:]:
+            per *= s
+        return n_blocks * L * 2 * per * self.pool.element_size()
 
     @staticmethod
     def _wire(t: torch.Tensor) -> torch.Tensor:
         """RCCL has no fp8 dtype: ship fp8 staging buffers as uint8."""
         return t.view(torch.uint8) if t.dtype == torch.float8_e4m3fn else t
 
-    def send_blocks(self, dst_rank: int, block_ids: List[int]) -> None:
-        t0 = time.monotonic()
+    def _finish(self, on_complete, nbytes: int, label: str,
+                staging=None) -> None:
+        if self.on_gpu:
+            with torch.cuda.stream(self.stream):
+                ev = torch.cuda.Event()
+                ev.record(self.stream)
+            self.pending.append(TransferHandle(ev, on_complete, nbytes,
+                                               label, staging))
+        else:
+            # CPU/gloo ops ran synchronously: complete immediately
+            TransferHandle(None, on_complete, nbytes, label).poll()
+
+    # ---- RCCL staged path ----
+    def send_blocks(self, dst_rank: int, block_ids: List[int],
+                    on_complete: Optional[Callable] = None) -> None:
         ids = torch.tensor(block_ids, dtype=torch.int32, device=self.device)
         staging = self._staging(len(block_ids))
         if self.on_gpu:
             with torch.cuda.stream(self.stream):
                 ops.move_blocks(self.pool, staging, ids, is_scatter=False)
                 dist.send(self._wire(staging), dst=dst_rank, group=self.group)
-            self.stream.synchronize()
         else:
             ops.move_blocks(self.pool, staging, ids, is_scatter=False)
             dist.send(self._wire(staging), dst=dst_rank, group=self.group)
-        nbytes = staging.numel() * staging.element_size()
-        prom.xgmi_kv_transfer_bytes.labels("send").inc(nbytes)
-        prom.xgmi_kv_transfer_seconds.observe(time.monotonic() - t0)
+        self._finish(on_complete, staging.numel() * staging.element_size(),
+                     "send", staging)
 
-    def recv_blocks(self, src_rank: int, block_ids: List[int]) -> None:
-        t0 = time.monotonic()
+    def recv_blocks(self, src_rank: int, block_ids: List[int],
+                    on_complete: Optional[Callable] = None) -> None:
         ids = torch.tensor(block_ids, dtype=torch.int32, device=self.device)
         staging = self._staging(len(block_ids))
         if self.on_gpu:
             with torch.cuda.stream(self.stream):
                 dist.recv(self._wire(staging), src=src_rank, group=self.group)
                 ops.move_blocks(self.pool, staging, ids, is_scatter=True)
-            self.stream.synchronize()
         else:
             dist.recv(self._wire(staging), src=src_rank, group=self.group)
             ops.move_blocks(self.pool, staging, ids, is_scatter=True)
-        nbytes = staging.numel() * staging.element_size()
-        prom.xgmi_kv_transfer_bytes.labels("recv").inc(nbytes)
-        prom.xgmi_kv_transfer_seconds.observe(time.monotonic() - t0)
+        self._finish(on_complete, staging.numel() * staging.element_size(),
+                     "recv", staging)
+
+    def recv_discard(self, src_rank: int, n_blocks: int,
+                     on_complete: Optional[Callable] = None) -> None:
+        """Keep the P2P pairing matched when the receiver cannot adopt
+        (kv_exhausted): receive into scratch and drop."""
+        scratch = self._staging(n_blocks)
+        if self.on_gpu:
+            with torch.cuda.stream(self.stream):
+                dist.recv(self._wire(scratch), src=src_rank, group=self.group)
+        else:
+            dist.recv(self._wire(scratch), src=src_rank, group=self.group)
+        self._finish(on_complete, scratch.numel() * scratch.element_size(),
+                     "recv_discard", scratch)
+
+    # ---- direct peer-pull path (HIP IPC over xGMI) ----
+    def pull_blocks(self, src_rank: int, src_ids: List[int],
+                    dst_ids: List[int],
+                    on_complete: Optional[Callable] = None) -> None:
+        """One-sided gather from the mapped peer pool into local blocks."""
+        ptr, src_nb = self.peer_pools[src_rank]
+        s = torch.tensor(src_ids, dtype=torch.int32, device=self.device)
+        d = torch.tensor(dst_ids, dtype=torch.int32, device=self.device)
+        with torch.cuda.stream(self.stream):
+            ops.hip_ops().copy_blocks_peer(ptr, self.pool, s, d, src_nb)
+        self._finish(on_complete, self._block_nbytes(len(src_ids)), "pull")
 
     # ---- embeddings (encode -> prefill hand-off, E/P/D) ----
     def send_tensor(self, dst_rank: int, t: torch.Tensor) -> None:

@@ -171,14 +171,22 @@ def main():
 
     use_gpu = torch.cuda.is_available() if args.device is None \
         else args.device.startswith("cuda")
-    device = (args.device or (f"cuda:{local_rank}" if use_gpu else "cpu"))
+    # oversubscribed rig: more ranks than visible GPUs (e.g. a 2-rank P/D
+    # topology exercised on a 1-GPU box) — all ranks share cuda:0, the
+    # process group is gloo (RCCL cannot host two ranks on one device) and
+    # KV moves via the HIP-IPC peer-pull transport, which needs no RCCL
+    n_vis = torch.cuda.device_count() if use_gpu else 0
+    same_dev = use_gpu and world > n_vis
+    device = (args.device or
+              ("cuda:0" if same_dev else
+               (f"cuda:{local_rank}" if use_gpu else "cpu")))
     dtype = torch.bfloat16 if use_gpu else torch.float32
 
     import torch.distributed as dist
     mailbox_group = transfer_group = None
     if world > 1:
-        backend = "nccl" if use_gpu else "gloo"
-        if use_gpu:
+        backend = "nccl" if use_gpu and not same_dev else "gloo"
+        if use_gpu and not same_dev:
             torch.cuda.set_device(local_rank)
         dist.init_process_group(backend)
         transfer_group = None  # default group (RCCL over xGMI)
@@ -189,6 +197,11 @@ def main():
     from llm_d_inference_scheduler_amd.node import NodeConfig, NodeRunner
     from llm_d_inference_scheduler_amd.flowcontrol import BandConfig
     from llm_d_inference_scheduler_amd.api.objectives import InferenceObjective
+
+    if same_dev:
+        # all ranks share one 288 GB GPU: shrink each rank's KV budget so
+        # world * (weights + KV + activations) fits
+        args.kv_gb = min(args.kv_gb, max(16.0, 288.0 / world - 48.0))
 
     model_cfg = LLAMA_3_8B if args.model == "llama-3-8b" else TINY_LLAMA
     if args.mode == "epd":

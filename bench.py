@@ -273,7 +273,17 @@ def main():
         except (ValueError, IndexError):
             return "standard"
 
-    ramp = max(1, target_inflight // max(8, args.warmup // 2))
+    # Pace closed-loop admission to the prefill service rate: one prefill
+    # pass is prefill_chunk_tokens (8192) per prompt-running rank, so
+    # feeding ~one pass worth of prompts per step keeps every arrival's
+    # individual TTFT at ~1 queued pass (well under the SLO) instead of
+    # building a cold-start backlog whose tail blows the 2s SLO and then
+    # pollutes goodput for the next `max_tokens` steps (slo_ok is frozen
+    # per request at first token).
+    from llm_d_inference_scheduler_amd.datalayer.endpoint import Role as _R
+    n_prompt_ranks = (len(node.topology.ranks_with(_R.PREFILL))
+                      or max(1, n_decode))
+    ramp = max(1, n_prompt_ranks * 8192 // max(1, args.prompt_len))
     arrival_state = {"next": None}
 
     def feed(limit=None):

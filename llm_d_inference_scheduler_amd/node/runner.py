@@ -906,6 +906,17 @@ class NodeRunner:
                 self.engine.admit_transferred(pending["req"], local,
                                               job["seq_len"],
                                               job["first_token"])
+                # true TTFT through the hand-off (connector_nixlv2.go
+                # true_ttft_ms span attr): first token is client-visible
+                # only once the decode side adopted the KV
+                req = pending["req"]
+                if req.arrival_t and req.first_token_t:
+                    ttft = (req.first_token_t - req.arrival_t) * 1e3
+                    if self.is_router:
+                        self._ttft_events.append(ttft)
+                    else:
+                        self._outbox.append({"type": "ttft",
+                                             "req_id": req_id, "ms": ttft})
             if peer:
                 self._outbox.append({"type": "kv_released",
                                      "req_id": req_id, "dst": src})

@@ -215,6 +215,13 @@ class ParserMux:
         parser = self.by_content_type.get(ct, self.default)
         return parser.parse_request(body, headers, path)
 
+    def parse_response_usage(self, body: bytes, headers: Dict[str, str],
+                             streaming: bool = False) -> Optional[Usage]:
+        ct = headers.get("content-type", "application/json").split(";")[0]
+        parser = self.by_content_type.get(ct, self.default)
+        streaming = streaming or body.lstrip().startswith(b"data:")
+        return parser.parse_response_usage(body, streaming)
+
 
 # ---------------------------------------------------------------------------
 # vLLM gRPC parser (parsers/vllmgrpc): the reference parses the vLLM

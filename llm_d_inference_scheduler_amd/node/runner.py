@@ -387,6 +387,8 @@ class NodeRunner:
         if not self.detector.is_saturated(self.datastore.endpoints()):
             return
         for rid in self.evictor.evict_n(1, lambda item: None):
+            if getattr(self, "extproc", None) is not None:
+                self.extproc.evict(rid)     # open ext-proc stream -> 429
             self._chunked.pop(rid, None)
             decision = self._decisions.pop(rid, None)
             self._outbox.append({"type": "abort", "req_id": rid})

@@ -36,6 +36,13 @@ class ShardProcessor:
         self._wake = threading.Event()
         self._stop = threading.Event()
         self._thread: Optional[threading.Thread] = None
+        # draining: no NEW work via JSQ; retires once fully drained
+        # (registry shard.go draining lifecycle)
+        self.draining = False
+
+    @property
+    def drained(self) -> bool:
+        return self.draining and self.queued_len == 0
 
     # ---- stats used by JSQ-bytes shard selection (controller.go) ----
     @property
@@ -72,16 +79,19 @@ class ShardProcessor:
                     return
             q.push(item)
             self.registry.record("enqueued")
+            self.registry.touch_flow(item.flow_key)
             prom.flow_queue_size.labels(str(bc.priority)).set(band.total_len)
         self._wake.set()
 
     def _over_global(self, extra_bytes: int) -> bool:
+        # this shard's slice of the partitioned global capacity (shard.go)
         r = self.registry
-        if r.global_max_items is not None and \
-                sum(b.total_len for b in self.bands.values()) + 1 > r.global_max_items:
+        max_items, max_bytes = r.shard_max_items, r.shard_max_bytes
+        if max_items is not None and \
+                sum(b.total_len for b in self.bands.values()) + 1 > max_items:
             return True
-        if r.global_max_bytes is not None and \
-                sum(b.total_bytes for b in self.bands.values()) + extra_bytes > r.global_max_bytes:
+        if max_bytes is not None and \
+                sum(b.total_bytes for b in self.bands.values()) + extra_bytes > max_bytes:
             return True
         return False
 
@@ -138,7 +148,7 @@ class ShardProcessor:
                         (time.monotonic_ns() - item.enqueue_ns) / 1e9)
                     dispatched += 1
             for band in self.bands.values():
-                band.gc()
+                band.gc(self.registry.keep_flow)
         return dispatched
 
     def _select_item(self):

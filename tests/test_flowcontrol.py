@@ -311,3 +311,114 @@ class TestCapacityBoundaries:
         z = mk_req("z")
         fc0.submit(z)
         assert z.outcome == QueueOutcome.REJECTED_CAPACITY
+
+
+class TestFlowLifecycle:
+    """Registry flow leasing + idle GC + elastic shard topology
+    (registry.go:79-310 leasing/connection/GC, shard.go draining)."""
+
+    def test_lease_keeps_empty_flow_alive(self):
+        reg = FlowRegistry(bands=[BandConfig(0)], flow_idle_ttl_s=0.0)
+        out = []
+        fc = FlowController(reg, collector(out))
+        reg.open_connection("tenant-a")
+        fc.submit(mk_req("r1", flow="tenant-a"))
+        fc.tick()                     # dispatches; queue drains
+        fc.tick()                     # gc pass
+        band = fc.shards[0].bands[0]
+        assert "tenant-a" in band.flows       # leased: queue retained
+        reg.close_connection("tenant-a")
+        fc.tick()
+        assert "tenant-a" not in band.flows   # leaseless + idle: GC'd
+        assert reg.gc_flows() >= 1            # lifecycle record dropped
+        assert "tenant-a" not in reg.flows
+
+    def test_wait_holds_lease_for_queue_residency(self):
+        reg = FlowRegistry(bands=[BandConfig(0)], flow_idle_ttl_s=0.0)
+        fc = FlowController(reg, lambda i: False)   # never dispatches
+        item = mk_req("r1", flow="t", ttl=0.2)
+        t = threading.Thread(
+            target=lambda: fc.enqueue_and_wait(item, timeout=0.3))
+        t.start()
+        time.sleep(0.05)
+        assert reg.flows["t"].leases == 1
+        t.join(2.0)
+        assert reg.flows["t"].leases == 0
+
+    def test_jsq_bytes_spreads_across_shards(self):
+        reg = FlowRegistry(bands=[BandConfig(0)], num_shards=2)
+        fc = FlowController(reg, lambda i: False)   # hold everything queued
+        for i in range(8):
+            fc.submit(mk_req(f"r{i}", size=10))
+        lens = sorted(s.queued_len for s in fc.shards)
+        assert lens == [4, 4]                       # balanced by JSQ
+
+    def test_global_capacity_partitioned_over_shards(self):
+        reg = FlowRegistry(bands=[BandConfig(0)], num_shards=2,
+                           global_max_items=4)
+        fc = FlowController(reg, lambda i: False)
+        outcomes = []
+        items = [mk_req(f"r{i}") for i in range(6)]
+        for it in items:
+            fc.submit(it)
+        # 4 queued (2/shard), 2 rejected at the partitioned cap
+        assert fc.queued_len == 4
+        rejected = [it for it in items
+                    if it.outcome == QueueOutcome.REJECTED_CAPACITY]
+        assert len(rejected) == 2
+
+    def test_shard_scale_up_down_drains_without_loss(self):
+        reg = FlowRegistry(bands=[BandConfig(0)], num_shards=2)
+        out = []
+        gate = {"open": False}
+        fc = FlowController(reg, lambda i: gate["open"] and
+                            (out.append(i) or True))
+        items = [mk_req(f"r{i}") for i in range(10)]
+        for it in items[:6]:
+            fc.submit(it)
+        fc.set_shard_count(1)                 # shard 1 drains
+        assert any(s.draining for s in fc.shards)
+        for it in items[6:]:
+            fc.submit(it)                     # new work -> shard 0 only
+        assert all(not s.draining or s.queued_len > 0
+                   for s in fc.shards)
+        gate["open"] = True
+        for _ in range(5):
+            fc.tick()
+        assert len(out) == 10                 # nothing lost in the drain
+        assert len(fc.shards) == 1            # drained shard reaped
+        assert not fc.shards[0].draining
+
+    def test_churn_flows_under_load(self):
+        """Adversarial churn: 50 flows appear, burst, and drain while
+        dispatch stalls intermittently; every item finalizes exactly once
+        and leaseless idle flows are collected."""
+        reg = FlowRegistry(bands=[BandConfig(10), BandConfig(0)],
+                           num_shards=3, flow_idle_ttl_s=0.0)
+        state = {"n": 0}
+
+        def flaky_dispatch(item):
+            state["n"] += 1
+            return state["n"] % 3 != 0        # stall every 3rd attempt
+        fc = FlowController(reg, flaky_dispatch)
+        items = []
+        for wave in range(5):
+            for f in range(10):
+                it = mk_req(f"w{wave}f{f}", flow=f"flow-{wave}-{f}",
+                            priority=10 if f % 2 else 0, size=1 + f)
+                items.append(it)
+                reg.open_connection(it.flow_key)
+                fc.submit(it)
+            fc.tick()
+            if wave % 2 == 0:
+                fc.set_shard_count(2 + wave % 3)
+        for _ in range(30):
+            if all(it.finalized for it in items):
+                break
+            fc.tick()
+        assert all(it.outcome == QueueOutcome.DISPATCHED for it in items)
+        for it in items:
+            reg.close_connection(it.flow_key)
+        fc.tick()
+        reg.gc_flows()
+        assert not reg.flows                   # all lifecycle records GC'd

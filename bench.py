@@ -191,6 +191,12 @@ def main():
         dist.init_process_group(backend)
         transfer_group = None  # default group (RCCL over xGMI)
         mailbox_group = dist.new_group(backend="gloo")
+        # bench-level control collectives get their OWN gloo group: the
+        # mailbox group now runs a pipelined exchange on a helper thread,
+        # and concurrent collectives on one gloo group are unordered
+        ctrl_group = dist.new_group(backend="gloo")
+    else:
+        ctrl_group = None
 
     from llm_d_inference_scheduler_amd.models.configs import (
         LLAMA_3_8B, LLAVA_1_5_7B_TEXT, TINY_LLAMA, TINY_LLAVA)
@@ -376,7 +382,7 @@ def main():
             if world > 1:
                 rb = torch.tensor([ready], dtype=torch.float64)
                 dist.all_reduce(rb, op=dist.ReduceOp.MIN,
-                                group=mailbox_group)
+                                group=ctrl_group)
                 ready = float(rb[0])
             if ready >= 1.0:
                 break
@@ -402,10 +408,10 @@ def main():
     # MAX elapsed over ranks; SUM tokens over ranks
     if world > 1:
         buf = torch.tensor([elapsed], dtype=torch.float64)
-        dist.all_reduce(buf, op=dist.ReduceOp.MAX, group=mailbox_group)
+        dist.all_reduce(buf, op=dist.ReduceOp.MAX, group=ctrl_group)
         elapsed = float(buf[0])
         tbuf = torch.tensor([tokens, tokens_slo], dtype=torch.float64)
-        dist.all_reduce(tbuf, op=dist.ReduceOp.SUM, group=mailbox_group)
+        dist.all_reduce(tbuf, op=dist.ReduceOp.SUM, group=ctrl_group)
         tokens, tokens_slo = float(tbuf[0]), float(tbuf[1])
 
     if rank == 0:

@@ -68,10 +68,22 @@ class EndpointMetadata:
     address: str                # "rank:port"-style locator; rank for in-node
     rank: int = 0               # torch.distributed rank of the worker
     labels: Dict[str, str] = field(default_factory=dict)
+    # role-mask cache keyed on the label VALUE (labels are mutable; the
+    # string parse in role_mask was 4 us x ~10 calls per routed request —
+    # profiles/router_tax.json)
+    _role_lbl: str = field(default="", repr=False, compare=False)
+    _role_mask: int = field(default=0, repr=False, compare=False)
+
+    def _mask(self) -> int:
+        lbl = self.labels.get(ROLE_LABEL, "decode")
+        if lbl != self._role_lbl:
+            self._role_mask = int(role_mask(lbl))
+            self._role_lbl = lbl
+        return self._role_mask
 
     @property
     def roles(self) -> Role:
-        return role_mask(self.labels.get(ROLE_LABEL, "decode"))
+        return Role(self._mask())
 
 
 class Endpoint:
@@ -117,7 +129,9 @@ class Endpoint:
         return self.metadata.roles
 
     def has_role(self, role: Role) -> bool:
-        return bool(self.metadata.roles & role)
+        # plain int AND — IntFlag.__and__ re-constructs an enum member per
+        # call and showed up at 40 us/request in the router profile
+        return bool(self.metadata._mask() & int(role))
 
     def __repr__(self) -> str:
         return f"Endpoint({self.metadata.name}, roles={self.metadata.roles!r})"

@@ -99,5 +99,19 @@ xgmi_kv_transfer_seconds = Histogram(
     buckets=(.0001, .00025, .0005, .001, .0025, .005, .01, .025, .05, .1))
 
 
+_label_cache = {}
+
+
+def L(metric, *vals):
+    """Memoized metric.labels(*vals): prometheus_client's labels() costs
+    ~3 us (lock + validation) per call — measurable on the per-request
+    routing path (profiles/router_tax.json). Use for hot-path metrics."""
+    key = (id(metric),) + vals
+    child = _label_cache.get(key)
+    if child is None:
+        child = _label_cache[key] = metric.labels(*vals)
+    return child
+
+
 def render() -> bytes:
     return generate_latest(registry)

@@ -93,6 +93,13 @@ class NodeConfig:
     # default as making it vacuous)
     fc_num_shards: int = 2
     route_batch_per_step: int = 64
+    # route OFF the lockstep thread (the reference's EPP runs beside, not
+    # inside, the serving loop): arrivals are handed to a worker pool and
+    # finished decisions drain each step, so the rank-0 step time is
+    # independent of the arrival rate (profiles/router_tax.md). None =
+    # auto (on when flow_control is on, where admission can block; off
+    # otherwise for bit-reproducible single-thread routing)
+    route_async: Optional[bool] = None
     ttft_slo_ms: Optional[float] = None
     # P/D stage choreography variant (reference sidecar connectors,
     # proxy.go:72-81), re-grounded on the xGMI transfer engine:
@@ -148,7 +155,13 @@ class NodeRunner:
         self.rank = cfg.rank
         self.topology = NodeTopology.parse(cfg.topology, cfg.world_size)
         self.my_spec = self.topology.ranks[self.rank]
-        self.mailbox = Mailbox(cfg.mailbox_group, self.rank, cfg.world_size)
+        # pipelined (one-step-deep) exchange only when the caller gave the
+        # mailbox its own gloo group: pipelining runs the collective on a
+        # helper thread, which is only safe if nothing else issues
+        # collectives on that group (two concurrent collectives on one
+        # gloo group are unordered)
+        self.mailbox = Mailbox(cfg.mailbox_group, self.rank, cfg.world_size,
+                               pipelined=cfg.mailbox_group is not None)
 
         want_ipc = (cfg.world_size > 1
                     and str(cfg.device).startswith("cuda")
@@ -339,11 +352,21 @@ class NodeRunner:
         self._assign_seq = 0
         from collections import deque
         self.epp_latencies = deque(maxlen=100_000)  # ms, per routed request
-        # flow-control mode: admission blocks in the queue, so routing runs
-        # on a pool and finished decisions drain into the outbox each step
-        self._route_pool = (ThreadPoolExecutor(max_workers=64,
+        # async routing pool: mandatory in flow-control mode (admission
+        # blocks in the queue), optional elsewhere (cfg.route_async) to
+        # decouple step time from arrival rate
+        use_pool = cfg.route_async if cfg.route_async is not None \
+            else self.flow is not None
+        # worker count: flow-control admission BLOCKS in the queue, so it
+        # needs enough threads to cover queued residency (64); plain async
+        # routing never blocks and more threads just thrash the GIL and
+        # the director's scheduling lock (measured 663 req/s at 64 threads
+        # vs ~5.8k single-thread — profiles/router_tax.json)
+        n_workers = 64 if self.flow is not None else 2
+        self._route_pool = (ThreadPoolExecutor(max_workers=n_workers,
                                                thread_name_prefix="route")
-                            if self.flow is not None else None)
+                            if (use_pool or self.flow is not None)
+                            else None)
         self._routed: "queue_mod.Queue" = queue_mod.Queue()
 
     # ------------------------------------------------------------------
@@ -988,6 +1011,7 @@ class NodeRunner:
     # ------------------------------------------------------------------
     def shutdown(self) -> None:
         self.transfer.synchronize()   # drain in-flight xGMI transfers
+        self.mailbox.drain()          # flush the pipelined exchange
         if self.is_router and self.flow is not None:
             self.flow.stop()
         if self.is_router and getattr(self, "remote", None) is not None:

@@ -17,7 +17,8 @@ class _RoleFilter(Filter):
     role: Role = Role.DECODE
 
     def filter(self, ctx, endpoints: List[Endpoint]) -> List[Endpoint]:
-        return [ep for ep in endpoints if ep.has_role(self.role)]
+        r = int(self.role)
+        return [ep for ep in endpoints if ep.metadata._mask() & r]
 
 
 @register_plugin("decode-filter", aliases=["decode_filter"])

@@ -164,8 +164,8 @@ class Director:
                                           status=503)
                 self._prepare_request(ctx, result, target)
             latency_ms = (time.monotonic() - t0) * 1e3
-            prom.request_total.labels(req.model, req.target_model).inc()
-            prom.running_requests.labels(req.target_model).inc()
+            prom.L(prom.request_total, req.model, req.target_model).inc()
+            prom.L(prom.running_requests, req.target_model).inc()
             return RoutingDecision(request=req, target=target, result=result,
                                    ctx=ctx, epp_latency_ms=latency_ms)
 
@@ -202,7 +202,7 @@ class Director:
                 log.v(3).info("data producer budget exhausted",
                               skipped=producer.name)
                 break
-            with prom.plugin_latency.labels(producer.name).time():
+            with prom.L(prom.plugin_latency, producer.name).time():
                 try:
                     producer.produce(ctx, endpoints)
                 except Exception as e:
@@ -218,7 +218,7 @@ class Director:
         if hasattr(handler, "pre_request"):
             handler.pre_request(ctx, result, target)
         for plugin in self.config.pre_request:
-            with prom.plugin_latency.labels(plugin.name).time():
+            with prom.L(prom.plugin_latency, plugin.name).time():
                 plugin.pre_request(ctx, result, target)
 
     # ---- response path (director.go:384 HandleResponseHeader,
@@ -237,7 +237,7 @@ class Director:
         req = decision.request
         for plugin in self.config.response_complete:
             plugin.response_complete(decision.ctx, decision.target, usage)
-        prom.running_requests.labels(req.target_model).dec()
+        prom.L(prom.running_requests, req.target_model).dec()
         if usage is not None:
             model = req.model
             if getattr(usage, "prompt_tokens", 0):

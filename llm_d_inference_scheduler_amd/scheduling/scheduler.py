@@ -68,7 +68,7 @@ class SchedulerProfile:
 
         extra = np.zeros(len(eps), dtype=np.float32)
         for scorer, weight in python_scorers:
-            with prom.plugin_latency.labels(scorer.name).time():
+            with prom.L(prom.plugin_latency, scorer.name).time():
                 smap = scorer.score(ctx, eps)
             for i, ep in enumerate(eps):
                 v = min(1.0, max(0.0, smap.get(ep.name, 0.0)))
@@ -110,7 +110,7 @@ def _snapshot_arrays(eps: List[Endpoint]) -> Dict[str, np.ndarray]:
     from ..datalayer.attributes import IN_FLIGHT_LOAD
     for i, ep in enumerate(eps):
         m = ep.metrics
-        snap["roles"][i] = int(ep.roles) & 0xFF
+        snap["roles"][i] = ep.metadata._mask() & 0xFF
         snap["queue"][i] = m.waiting_queue_size
         snap["running"][i] = m.running_requests_size
         snap["kv"][i] = m.kv_cache_usage

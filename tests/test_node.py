@@ -369,3 +369,59 @@ class TestTransferAsyncUnit:
         captured["cb"]()                      # transfer completes
         assert mgr.free_blocks == free0       # released exactly once
         node.shutdown()
+
+
+def _pd_pipelined_worker(rank, world_size, init_file, out_file):
+    """P/D with the PIPELINED mailbox (dedicated gloo subgroup, one-step-
+    deep exchange overlap — the production bench configuration)."""
+    import torch.distributed as dist
+    dist.init_process_group("gloo", init_method=f"file://{init_file}",
+                            rank=rank, world_size=world_size)
+    mailbox_group = dist.new_group(backend="gloo")
+    ctrl_group = dist.new_group(backend="gloo")
+    try:
+        cfg = NodeConfig(model=TINY_LLAMA, rank=rank, world_size=world_size,
+                         topology="pd:1p1d", device="cpu",
+                         dtype=torch.float32, kv_blocks=256,
+                         epp_yaml=EPP_YAML_TIGHT_DISAGG, seed=3,
+                         mailbox_group=mailbox_group)
+        node = NodeRunner(cfg)
+        assert node.mailbox.pipelined
+        results = []
+        if rank == 0:
+            for i in range(5):
+                node.submit(make_req(i, n_prompt=48, max_tokens=4))
+        for _ in range(300):
+            node.step()
+            if rank == 0:
+                results.extend(node.drain_completions())
+                done = torch.tensor([1 if len(results) >= 5 else 0])
+            else:
+                done = torch.tensor([0])
+            dist.broadcast(done, src=0, group=ctrl_group)
+            if done.item():
+                break
+        if rank == 0:
+            with open(out_file, "w") as f:
+                json.dump([{ "id": c.request_id, "error": c.error,
+                             "completion": c.usage.completion_tokens}
+                           for c in results], f)
+        node.shutdown()
+        dist.barrier(group=ctrl_group)
+    finally:
+        dist.destroy_process_group()
+
+
+@pytest.mark.timeout(240)
+class TestPipelinedMailbox:
+    def test_pd_with_pipelined_exchange(self, tmp_path):
+        init_file = str(tmp_path / "pgpl_init")
+        out_file = str(tmp_path / "outpl.json")
+        mp.start_processes(_pd_pipelined_worker,
+                           args=(2, init_file, out_file),
+                           nprocs=2, join=True, start_method="spawn")
+        with open(out_file) as f:
+            results = json.load(f)
+        assert len(results) == 5
+        assert all(not r["error"] for r in results), results
+        assert all(r["completion"] == 4 for r in results)

@@ -35,6 +35,7 @@ hip_ops = cpp_extension.CUDAExtension(
         "csrc/hip/norm_rope_act.hip",
         "csrc/hip/kv_cache.hip",
         "csrc/hip/paged_attention.hip",
+        "csrc/hip/paged_attention_v4.hip",
         "csrc/hip/flash_prefill.hip",
         "csrc/hip/flash_prefill_glds.hip",
     ],

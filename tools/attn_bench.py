@@ -1,5 +1,7 @@
 """Decode paged-attention kernel microbenchmark (run on a GPU box).
-Times the production kernel at serving shapes with/without sequence split.
+Times the production kernel at serving shapes with/without sequence split,
+against the pipelined v4 (producer/consumer waves, double-buffered logits).
+Also checks v4 numerics against v1 before timing.
 """
 import os
 import sys
@@ -13,25 +15,49 @@ import torch  # noqa: E402
 from llm_d_inference_scheduler_amd.ops import hip_ops  # noqa: E402
 
 ext = hip_ops()
-KVH, D, BS, QPG = 8, 128, 16, 4
-QH = KVH * QPG
+KVH, D, BS = 8, 128, 16
 
 
-def bench(B, ctx, np_=None, part=512, iters=50, v3=False):
-    torch.manual_seed(0)
+def make(B, ctx, qpg, fp8=False, seed=0):
+    torch.manual_seed(seed)
     max_blocks = (ctx + BS - 1) // BS
     NB = max_blocks * B + 1
-    q = torch.randn(B, QH, D, device="cuda").bfloat16()
+    q = torch.randn(B, KVH * qpg, D, device="cuda").bfloat16()
     kc = torch.randn(NB, KVH, BS, D, device="cuda").bfloat16()
     vc = torch.randn(NB, KVH, BS, D, device="cuda").bfloat16()
+    if fp8:
+        kc = kc.to(torch.float8_e4m3fn)
+        vc = vc.to(torch.float8_e4m3fn)
     bt = torch.arange(1, NB, dtype=torch.int32,
                       device="cuda").view(B, max_blocks)
     sl = torch.full((B,), ctx, dtype=torch.int32, device="cuda")
+    return q, kc, vc, bt, sl
+
+
+def check_v4(qpg=4, fp8=False):
+    for B, ctx, np_, part in ((4, 300, 1, 0), (8, 1152, 1, 0),
+                              (8, 1152, 4, 320), (3, 70, 1, 0)):
+        q, kc, vc, bt, sl = make(B, ctx, qpg, fp8)
+        scale = D ** -0.5
+        ref = ext.paged_attention(q, kc, vc, bt, sl, scale)
+        got = ext.paged_attention_v4(q, kc, vc, bt, sl, np_,
+                                     part if np_ > 1 else ctx + 256, scale)
+        diff = (ref.float() - got.float()).abs().max().item()
+        status = "OK" if diff < 3e-2 else "FAIL"
+        print(f"v4 numerics qpg={qpg} fp8={int(fp8)} B={B} ctx={ctx} "
+              f"np={np_}: max|d|={diff:.4f} {status}")
+
+
+def bench(B, ctx, np_=None, part=512, iters=50, kernel="v1", qpg=4):
+    q, kc, vc, bt, sl = make(B, ctx, qpg)
     scale = D ** -0.5
 
     def run():
-        if v3:
+        if kernel == "v3":
             return ext.paged_attention_v3(q, kc, vc, bt, sl, np_ or 1,
+                                          part if np_ else ctx + 256, scale)
+        if kernel == "v4":
+            return ext.paged_attention_v4(q, kc, vc, bt, sl, np_ or 1,
                                           part if np_ else ctx + 256, scale)
         if np_:
             return ext.paged_attention_split(q, kc, vc, bt, sl, np_, part,
@@ -47,23 +73,22 @@ def bench(B, ctx, np_=None, part=512, iters=50, v3=False):
     torch.cuda.synchronize()
     us = (time.perf_counter() - t0) / iters * 1e6
     kv_gb = B * ctx * KVH * D * 2 * 2 / 1e9
-    tag = " v3" if v3 else "   "
-    print(f"B={B:4d} ctx={ctx:5d} np={np_ or 1:2d}{tag}: {us:8.1f} us  "
+    print(f"B={B:4d} ctx={ctx:5d} np={np_ or 1:2d} {kernel}: {us:8.1f} us  "
           f"({kv_gb / (us / 1e6) / 1e3:6.2f} TB/s effective)")
 
 
 if __name__ == "__main__":
+    check_v4(qpg=4)
+    check_v4(qpg=4, fp8=True)
+    check_v4(qpg=8)
     for B in (64, 128, 256):
         bench(B, 1152)
-        bench(B, 1152, v3=True)
-        for np_ in (2, 4, 8):
-            part = ((1152 + np_ - 1) // np_ + 63) // 64 * 64
+        bench(B, 1152, kernel="v4")
+        for np_ in (2, 4):
+            part = ((1152 + np_ - 1) // np_ + 255) // 256 * 256
             bench(B, 1152, np_=np_, part=part)
-            bench(B, 1152, np_=np_, part=part, v3=True)
+            bench(B, 1152, np_=np_, part=part, kernel="v4")
     bench(8, 8192, np_=16, part=512)
-    bench(8, 8192, np_=16, part=512, v3=True)
+    bench(8, 8192, np_=16, part=512, kernel="v4")
     bench(128, 1536)
-    bench(128, 1536, v3=True)
-    for np_ in (2, 4):
-        part = ((1536 + np_ - 1) // np_ + 63) // 64 * 64
-        bench(128, 1536, np_=np_, part=part, v3=True)
+    bench(128, 1536, kernel="v4")

@@ -275,7 +275,7 @@ class NodeRunner:
         from ..models.tokenizer import HashTokenizer
         from ..plugins.producers import TokenProducer
         for p in self.loaded.plugins.values():
-            if isinstance(p, TokenProducer):
+            if isinstance(p, TokenProducer) and p.mode == "inprocess":
                 p.tokenizer = HashTokenizer(cfg.model.vocab_size)
         self.detector = UtilizationSaturationDetector()
         candidates = EndpointCandidates(self.datastore, cache_ttl_s=0.0)

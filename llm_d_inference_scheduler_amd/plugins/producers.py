@@ -45,9 +45,22 @@ class TokenProducer(DataProducer):
     def __init__(self, name: str = "", **params):
         super().__init__(name, **params)
         self.tokenizer = params.get("tokenizer")
+        # remote-tokenizer surface (tokenizer/{vllm_http,uds}.go): mode
+        #   inprocess (default) — shared hash tokenizer
+        #   http  — POST <url>/tokenize on a vLLM-compatible worker
+        #   uds   — same HTTP over a unix socket (the UDS sidecar analog)
+        self.mode = params.get("mode", "inprocess")
         if self.tokenizer is None:
-            from ..models.tokenizer import HashTokenizer
-            self.tokenizer = HashTokenizer()
+            if self.mode in ("http", "uds"):
+                from ..models.tokenizer import HttpTokenizer
+                self.tokenizer = HttpTokenizer(
+                    base_url=params.get("url", ""),
+                    uds_path=params.get("udsPath", ""),
+                    model=params.get("model", ""),
+                    timeout_s=float(params.get("timeoutMs", 400)) / 1e3)
+            else:
+                from ..models.tokenizer import HashTokenizer
+                self.tokenizer = HashTokenizer()
 
     def produce(self, ctx: SchedulingContext, endpoints) -> None:
         req = ctx.request

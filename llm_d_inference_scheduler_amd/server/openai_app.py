@@ -197,6 +197,22 @@ def build_app(service: NodeService,
             {"id": service.node.cfg.model.name, "object": "model",
              "owned_by": "llm-d-inference-scheduler-amd"}]}
 
+    @app.post("/tokenize")
+    async def tokenize(request: Request):
+        """vLLM-compatible tokenize route: lets this node serve as the
+        remote-tokenizer worker for a peer's token producer (the surface
+        the reference consumes via dataproducer/tokenizer/vllm_http.go)."""
+        body = await request.json()
+        toks = tok(str(body.get("prompt", "")))
+        return {"tokens": toks, "count": len(toks),
+                "max_model_len": service.node.engine.max_model_len}
+
+    @app.post("/detokenize")
+    async def detokenize(request: Request):
+        body = await request.json()
+        return {"prompt": tok.decode([int(t)
+                                      for t in body.get("tokens", [])])}
+
     @app.get("/metrics")
     async def metrics():
         return Response(content=prom.render() + _vllm_compat(service),

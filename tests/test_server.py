@@ -252,3 +252,70 @@ class TestVllmCompatScrape:
         assert m is not None
         assert m.waiting_queue_size >= 0 and 0.0 <= m.kv_cache_usage <= 1.0
         assert m.cache_num_blocks > 0      # cache_config_info present
+
+
+class TestTokenizerSurface:
+    """Remote-tokenizer surface (reference dataproducer/tokenizer/
+    {vllm_http,uds}.go): /tokenize + /detokenize routes and the
+    HttpTokenizer client with fail-open fallback."""
+
+    def _client(self):
+        cfg = NodeConfig(model=TINY_LLAMA, world_size=1, topology="mono",
+                         device="cpu", dtype=torch.float32, kv_blocks=64)
+        service = NodeService(NodeRunner(cfg))
+        service.start()
+        return TestClient(build_app(service)), service
+
+    def test_tokenize_route_roundtrip(self):
+        c, service = self._client()
+        with c:
+            r = c.post("/tokenize", json={"model": "tiny-llama",
+                                          "prompt": "hello world, again"})
+            assert r.status_code == 200
+            body = r.json()
+            assert body["count"] == len(body["tokens"]) > 0
+            r2 = c.post("/detokenize", json={"tokens": body["tokens"]})
+            assert r2.status_code == 200 and r2.json()["prompt"]
+        service.stop()
+
+    def test_http_tokenizer_against_front_door(self):
+        from llm_d_inference_scheduler_amd.models.tokenizer import (
+            HashTokenizer, HttpTokenizer)
+        c, service = self._client()
+        with c:
+            tok = HttpTokenizer(model="tiny-llama", client=c,
+                                fallback=HashTokenizer(1000))
+            ids = tok("the quick brown fox")
+            # the server used its own HashTokenizer: results must agree
+            assert ids == HashTokenizer()("the quick brown fox")
+            assert tok.errors == 0
+        service.stop()
+
+    def test_http_tokenizer_fail_open(self):
+        from llm_d_inference_scheduler_amd.models.tokenizer import (
+            HashTokenizer, HttpTokenizer)
+
+        class DeadClient:
+            def post(self, *a, **k):
+                raise ConnectionError("worker down")
+        tok = HttpTokenizer(model="m", client=DeadClient(),
+                            fallback=HashTokenizer(5000))
+        ids = tok("still routes")
+        assert ids == HashTokenizer(5000)("still routes")
+        assert tok.errors == 1
+
+    def test_token_producer_http_mode(self):
+        from llm_d_inference_scheduler_amd.plugins.producers import (
+            TokenProducer)
+        from llm_d_inference_scheduler_amd.models.tokenizer import (
+            HttpTokenizer)
+        p = TokenProducer("token-producer", mode="http",
+                          url="http://127.0.0.1:1", timeoutMs=50)
+        assert isinstance(p.tokenizer, HttpTokenizer)
+        # unreachable worker -> fail-open to the hash fallback
+        from llm_d_inference_scheduler_amd.scheduling.types import (
+            LLMRequest, SchedulingContext)
+        req = LLMRequest(request_id="t", model="m", prompt="abc def")
+        ctx = SchedulingContext(request=req)
+        p.produce(ctx, [])
+        assert req.prompt_tokens and p.tokenizer.errors == 1

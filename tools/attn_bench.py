@@ -34,17 +34,18 @@ def make(B, ctx, qpg, fp8=False, seed=0):
     return q, kc, vc, bt, sl
 
 
-def check_v4(qpg=4, fp8=False):
+def check(kernel, qpg=4, fp8=False):
+    fn = getattr(ext, f"paged_attention_{kernel}")
     for B, ctx, np_, part in ((4, 300, 1, 0), (8, 1152, 1, 0),
                               (8, 1152, 4, 320), (3, 70, 1, 0)):
         q, kc, vc, bt, sl = make(B, ctx, qpg, fp8)
         scale = D ** -0.5
         ref = ext.paged_attention(q, kc, vc, bt, sl, scale)
-        got = ext.paged_attention_v4(q, kc, vc, bt, sl, np_,
-                                     part if np_ > 1 else ctx + 256, scale)
+        got = fn(q, kc, vc, bt, sl, np_,
+                 part if np_ > 1 else ctx + 256, scale)
         diff = (ref.float() - got.float()).abs().max().item()
         status = "OK" if diff < 3e-2 else "FAIL"
-        print(f"v4 numerics qpg={qpg} fp8={int(fp8)} B={B} ctx={ctx} "
+        print(f"{kernel} numerics qpg={qpg} fp8={int(fp8)} B={B} ctx={ctx} "
               f"np={np_}: max|d|={diff:.4f} {status}")
 
 
@@ -58,6 +59,9 @@ def bench(B, ctx, np_=None, part=512, iters=50, kernel="v1", qpg=4):
                                           part if np_ else ctx + 256, scale)
         if kernel == "v4":
             return ext.paged_attention_v4(q, kc, vc, bt, sl, np_ or 1,
+                                          part if np_ else ctx + 256, scale)
+        if kernel == "v5":
+            return ext.paged_attention_v5(q, kc, vc, bt, sl, np_ or 1,
                                           part if np_ else ctx + 256, scale)
         if np_:
             return ext.paged_attention_split(q, kc, vc, bt, sl, np_, part,
@@ -78,17 +82,19 @@ def bench(B, ctx, np_=None, part=512, iters=50, kernel="v1", qpg=4):
 
 
 if __name__ == "__main__":
-    check_v4(qpg=4)
-    check_v4(qpg=4, fp8=True)
-    check_v4(qpg=8)
+    check("v5", qpg=4)
+    check("v5", qpg=4, fp8=True)
+    check("v5", qpg=8)
+    check("v5", qpg=1)
     for B in (64, 128, 256):
         bench(B, 1152)
-        bench(B, 1152, kernel="v4")
+        bench(B, 1152, kernel="v5")
         for np_ in (2, 4):
             part = ((1152 + np_ - 1) // np_ + 255) // 256 * 256
             bench(B, 1152, np_=np_, part=part)
-            bench(B, 1152, np_=np_, part=part, kernel="v4")
+            bench(B, 1152, np_=np_, part=part, kernel="v5")
     bench(8, 8192, np_=16, part=512)
-    bench(8, 8192, np_=16, part=512, kernel="v4")
+    bench(8, 8192, np_=16, part=512, kernel="v5")
     bench(128, 1536)
-    bench(128, 1536, kernel="v4")
+    bench(128, 1536, kernel="v5")
+    bench(256, 1152, kernel="v4")

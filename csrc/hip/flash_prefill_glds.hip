@@ -38,8 +38,22 @@ typedef __attribute__((ext_vector_type(16))) float f32x16;
 typedef __attribute__((ext_vector_type(4))) unsigned int uint4v;
 typedef __attribute__((ext_vector_type(4))) short short4v;
 
+// Bank swizzle S(row): swap the two 2-bit halves of row&15. Chosen from
+// the CDNA4 bank model (MI355X_MICROARCH.md "LDS"): the K ds_read_b128
+// only needs S bijective on row&15 (its 16-lane groups cover all 16
+// values of qcol&15, so any bijection spreads the 16 slots); the V
+// ds_read_b64_tr_b16 (32-lane groups) additionally needs kv0&3 steered
+// into address bits [7:6] so the 8-byte granule index
+// (dt | g | r) ^ (S << 1) is injective over the group's (g, kv0, r) —
+// the previous S(row)=row&15 left granule bits [4:3] untouched and put
+// FOUR lanes on every granule (the measured 47%-of-LDS-active conflict
+// cycles, profiles/r01_pmc_final.md).
+__device__ __forceinline__ int swz4(int row) {
+  return ((row & 3) << 2) | ((row >> 2) & 3);
+}
+
 __device__ __forceinline__ int g_swz(int row, int byte_off) {
-  return row * 256 + (byte_off ^ ((row & 15) << 4));
+  return row * 256 + (byte_off ^ (swz4(row) << 4));
 }
 
 __device__ __forceinline__ unsigned int pack2_bf16(float a, float b) {
@@ -122,7 +136,7 @@ __global__ __launch_bounds__(NW * WAVE, 2) void flash_prefill_glds_kernel(
     for (int j = 0; j < GLDS_PER_WAVE; ++j) {
       const int piece = (wave * GLDS_PER_WAVE + j) * 64 + lane;
       const int row = piece / 16, c16 = piece % 16;
-      const int src16 = c16 ^ (row & 15);       // source-side swizzle
+      const int src16 = c16 ^ swz4(row);        // source-side swizzle
       const int t = min(tile0 + row, ctx - 1);  // clamp: no garbage bf16
       const int64_t base =
           (((int64_t)bt_lds[t / bs] * kvh + kh) * bs + t % bs) * D;
@@ -232,12 +246,12 @@ __global__ __launch_bounds__(NW * WAVE, 2) void flash_prefill_glds_kernel(
             short4v lo = __builtin_amdgcn_ds_read_tr16_b64_v4i16(
                 (__attribute__((address_space(3))) short4v*)(
                     (const char*)v_lds + kv0 * 256 +
-                    (dc ^ ((kv0 & 15) << 4))));
+                    (dc ^ (swz4(kv0) << 4))));
             const int kv1 = kv0 + 4;
             short4v hi4 = __builtin_amdgcn_ds_read_tr16_b64_v4i16(
                 (__attribute__((address_space(3))) short4v*)(
                     (const char*)v_lds + kv1 * 256 +
-                    (dc ^ ((kv1 & 15) << 4))));
+                    (dc ^ (swz4(kv1) << 4))));
             short8 vfrag8 = {lo[0], lo[1], lo[2], lo[3],
                              hi4[0], hi4[1], hi4[2], hi4[3]};
             acc_o[dt] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(

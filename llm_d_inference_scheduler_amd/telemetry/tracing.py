@@ -22,8 +22,10 @@ class Span:
     start_ns: int
     end_ns: int = 0
     attributes: Dict[str, Any] = field(default_factory=dict)
-    parent: Optional[str] = None
+    parent: Optional[str] = None        # parent span NAME (in-process view)
     trace_id: str = ""
+    span_id: str = ""
+    parent_span_id: str = ""
 
     def set_attribute(self, key: str, value: Any) -> None:
         self.attributes[key] = value
@@ -46,10 +48,17 @@ class Tracer:
         self._spans: deque = deque(maxlen=capacity)
         self._lock = threading.Lock()
         self._local = threading.local()
+        self.otlp = None
         self.enabled = os.environ.get("LLMD_TRACING", "1") != "0"
         export = os.environ.get("LLMD_TRACE_EXPORT", "")
         if export and self.enabled:
             atexit.register(self.export_jsonl, export)
+        # OTLP/HTTP+JSON export (the reference's OTLP exporter,
+        # pkg/telemetry/tracing.go, env-driven): standard OTEL endpoint
+        # variable; spans batch on a daemon thread to <endpoint>/v1/traces
+        self.otlp = OtlpHttpExporter.from_env(service_name)
+        if self.otlp is not None and self.enabled:
+            atexit.register(self.otlp.flush)
 
     @contextmanager
     def span(self, name: str, **attrs):
@@ -57,10 +66,14 @@ class Tracer:
             yield _NOOP_SPAN
             return
         parent = getattr(self._local, "current", None)
+        import os
         s = Span(name=name, start_ns=time.monotonic_ns(),
                  attributes=dict(attrs),
                  parent=parent.name if parent else None,
-                 trace_id=parent.trace_id if parent else "")
+                 trace_id=parent.trace_id if parent else
+                 os.urandom(16).hex(),
+                 span_id=os.urandom(8).hex(),
+                 parent_span_id=parent.span_id if parent else "")
         prev = parent
         self._local.current = s
         try:
@@ -70,6 +83,8 @@ class Tracer:
             self._local.current = prev
             with self._lock:
                 self._spans.append(s)
+            if self.otlp is not None:
+                self.otlp.enqueue(s)
 
     def finished_spans(self, name: Optional[str] = None) -> List[Span]:
         with self._lock:
@@ -88,6 +103,101 @@ class Tracer:
                     "duration_ms": s.duration_ms, "parent": s.parent,
                     "trace_id": s.trace_id,
                     "attributes": s.attributes}) + "\n")
+
+
+class OtlpHttpExporter:
+    """Minimal OTLP/HTTP+JSON trace exporter (the protocol's official JSON
+    encoding of ExportTraceServiceRequest). Batches spans and POSTs to
+    `<OTEL_EXPORTER_OTLP_ENDPOINT>/v1/traces` on a daemon thread; drops on
+    transport failure (tracing must never block serving)."""
+
+    def __init__(self, endpoint: str, service_name: str,
+                 batch: int = 256, interval_s: float = 2.0, client=None):
+        self.endpoint = endpoint.rstrip("/")
+        self.service_name = service_name
+        self.batch = batch
+        self.interval_s = interval_s
+        self._buf: List[Span] = []
+        self._lock = threading.Lock()
+        self._client = client
+        self.sent = 0
+        self.dropped = 0
+        self._timer: Optional[threading.Timer] = None
+
+    @classmethod
+    def from_env(cls, service_name: str):
+        import os
+        ep = os.environ.get("OTEL_EXPORTER_OTLP_ENDPOINT", "")
+        if not ep:
+            return None
+        return cls(ep, service_name)
+
+    def enqueue(self, span: Span) -> None:
+        with self._lock:
+            self._buf.append(span)
+            full = len(self._buf) >= self.batch
+        if full:
+            self.flush()
+        elif self._timer is None:
+            t = threading.Timer(self.interval_s, self.flush)
+            t.daemon = True
+            self._timer = t
+            t.start()
+
+    @staticmethod
+    def _attr(k: str, v: Any) -> Dict[str, Any]:
+        if isinstance(v, bool):
+            val = {"boolValue": v}
+        elif isinstance(v, int):
+            val = {"intValue": str(v)}
+        elif isinstance(v, float):
+            val = {"doubleValue": v}
+        else:
+            val = {"stringValue": str(v)}
+        return {"key": k, "value": val}
+
+    def _payload(self, spans: List[Span]) -> Dict[str, Any]:
+        return {"resourceSpans": [{
+            "resource": {"attributes": [
+                self._attr("service.name", self.service_name)]},
+            "scopeSpans": [{
+                "scope": {"name": "llm_d_inference_scheduler_amd"},
+                "spans": [{
+                    "traceId": s.trace_id or "0" * 32,
+                    "spanId": s.span_id or "0" * 16,
+                    **({"parentSpanId": s.parent_span_id}
+                       if s.parent_span_id else {}),
+                    "name": s.name,
+                    "kind": 1,  # SPAN_KIND_INTERNAL
+                    "startTimeUnixNano": str(s.start_ns),
+                    "endTimeUnixNano": str(s.end_ns),
+                    "attributes": [self._attr(k, v)
+                                   for k, v in s.attributes.items()],
+                } for s in spans],
+            }],
+        }]}
+
+    def flush(self) -> None:
+        with self._lock:
+            spans, self._buf = self._buf, []
+            if self._timer is not None:
+                self._timer.cancel()
+                self._timer = None
+        if not spans:
+            return
+        try:
+            client = self._client
+            if client is None:
+                import httpx
+                client = self._client = httpx.Client(timeout=2.0)
+            r = client.post(self.endpoint + "/v1/traces",
+                            json=self._payload(spans))
+            if r.status_code < 300:
+                self.sent += len(spans)
+            else:
+                self.dropped += len(spans)
+        except Exception:
+            self.dropped += len(spans)
 
 
 class _NoopSpan:

@@ -253,3 +253,60 @@ class TestTraceparent:
         assert spans["gateway.request"].trace_id == tid
         assert spans["scheduler.schedule"].trace_id == tid
         assert spans["scheduler.schedule"].parent == "gateway.request"
+
+
+class TestOtlpExport:
+    """OTLP/HTTP+JSON trace export (pkg/telemetry/tracing.go analog),
+    env-gated via OTEL_EXPORTER_OTLP_ENDPOINT."""
+
+    def test_payload_shape_and_batching(self):
+        from llm_d_inference_scheduler_amd.telemetry.tracing import (
+            OtlpHttpExporter, Tracer)
+
+        posts = []
+
+        class FakeResp:
+            status_code = 200
+
+        class FakeClient:
+            def post(self, url, json=None):
+                posts.append((url, json))
+                return FakeResp()
+
+        tr = Tracer("svc-test")
+        tr.otlp = OtlpHttpExporter("http://collector:4318", "svc-test",
+                                   batch=2, client=FakeClient())
+        with tr.span("outer", model="m") as outer:
+            with tr.span("inner"):
+                pass
+        tr.otlp.flush()
+        assert tr.otlp.sent == 2
+        url, body = posts[-1]
+        assert url == "http://collector:4318/v1/traces"
+        rs = body["resourceSpans"][0]
+        svc = rs["resource"]["attributes"][0]
+        assert svc["key"] == "service.name"
+        spans = rs["scopeSpans"][0]["spans"]
+        by_name = {s["name"]: s for s in spans}
+        assert set(by_name) <= {"inner", "outer"}
+        inner = by_name["inner"]
+        assert len(inner["traceId"]) == 32 and len(inner["spanId"]) == 16
+        # child carries its parent's span id and trace id
+        if "outer" in by_name:
+            assert inner["parentSpanId"] == by_name["outer"]["spanId"]
+            assert inner["traceId"] == by_name["outer"]["traceId"]
+        assert int(inner["endTimeUnixNano"]) >= int(
+            inner["startTimeUnixNano"])
+
+    def test_fail_open_on_collector_down(self):
+        from llm_d_inference_scheduler_amd.telemetry.tracing import (
+            OtlpHttpExporter, Span)
+
+        class DeadClient:
+            def post(self, *a, **k):
+                raise ConnectionError("collector down")
+        ex = OtlpHttpExporter("http://dead:4318", "svc", client=DeadClient())
+        ex.enqueue(Span(name="s", start_ns=1, end_ns=2, trace_id="a" * 32,
+                        span_id="b" * 16))
+        ex.flush()
+        assert ex.dropped == 1 and ex.sent == 0

@@ -40,18 +40,6 @@ hipError_t lds_paged_attention_split(const void*, const void*, const void*,
                                      const int32_t*, const int32_t*, void*,
                                      float*, float*, int, int, int, int, int,
                                      int, int, int, int, float, hipStream_t);
-hipError_t lds_paged_attention_v3(const void*, const void*, const void*,
-                                  const int32_t*, const int32_t*, void*,
-                                  float*, float*, int, int, int, int, int,
-                                  int, int, int, int, float, hipStream_t);
-hipError_t lds_paged_attention_v4(const void*, const void*, const void*,
-                                  const int32_t*, const int32_t*, void*,
-                                  float*, float*, int, int, int, int, int,
-                                  int, int, int, int, float, hipStream_t);
-hipError_t lds_paged_attention_v5(const void*, const void*, const void*,
-                                  const int32_t*, const int32_t*, void*,
-                                  float*, float*, int, int, int, int, int,
-                                  int, int, int, int, float, hipStream_t);
 hipError_t lds_flash_prefill(const void*, const void*, const void*,
                              const int32_t*, const int32_t*, const int32_t*,
                              void*, int, int, int, int, int, int, int, float,
@@ -288,93 +276,6 @@ torch::Tensor paged_attention_split(torch::Tensor q, torch::Tensor k_cache,
   return out;
 }
 
-torch::Tensor paged_attention_v3(torch::Tensor q, torch::Tensor k_cache,
-                                 torch::Tensor v_cache,
-                                 torch::Tensor block_tables,
-                                 torch::Tensor seq_lens, int64_t n_parts,
-                                 int64_t part_tokens, double scale) {
-  CHECK_GPU(q); CHECK_GPU(k_cache); CHECK_GPU(v_cache);
-  CHECK_GPU(block_tables); CHECK_GPU(seq_lens);
-  int B = (int)q.size(0), qh = (int)q.size(1), d = (int)q.size(2);
-  int kvh = (int)k_cache.size(1), bs = (int)k_cache.size(2);
-  int qpg = qh / kvh;
-  auto out = torch::empty_like(q);
-  auto f32 = q.options().dtype(torch::kFloat32);
-  torch::Tensor part_o, part_ml;
-  float *po = nullptr, *pml = nullptr;
-  if (n_parts > 1) {
-    part_o = torch::empty({B, kvh, n_parts, qpg, d}, f32);
-    part_ml = torch::empty({B, kvh, n_parts, qpg, 2}, f32);
-    po = part_o.data_ptr<float>();
-    pml = part_ml.data_ptr<float>();
-  }
-  CHECK_HIP(lds_paged_attention_v3(
-      q.data_ptr(), k_cache.data_ptr(), v_cache.data_ptr(),
-      block_tables.data_ptr<int32_t>(), seq_lens.data_ptr<int32_t>(),
-      out.data_ptr(), po, pml, B, qh, kvh, bs, d,
-      (int)block_tables.size(1), (int)n_parts, (int)part_tokens,
-      kv_fp8_flag(k_cache), (float)scale, cur_stream()));
-  return out;
-}
-
-torch::Tensor paged_attention_v4(torch::Tensor q, torch::Tensor k_cache,
-                                 torch::Tensor v_cache,
-                                 torch::Tensor block_tables,
-                                 torch::Tensor seq_lens, int64_t n_parts,
-                                 int64_t part_tokens, double scale) {
-  CHECK_GPU(q); CHECK_GPU(k_cache); CHECK_GPU(v_cache);
-  CHECK_GPU(block_tables); CHECK_GPU(seq_lens);
-  int B = (int)q.size(0), qh = (int)q.size(1), d = (int)q.size(2);
-  int kvh = (int)k_cache.size(1), bs = (int)k_cache.size(2);
-  int qpg = qh / kvh;
-  auto out = torch::empty_like(q);
-  auto f32 = q.options().dtype(torch::kFloat32);
-  torch::Tensor part_o, part_ml;
-  float *po = nullptr, *pml = nullptr;
-  if (n_parts > 1) {
-    part_o = torch::empty({B, kvh, n_parts, qpg, d}, f32);
-    part_ml = torch::empty({B, kvh, n_parts, qpg, 2}, f32);
-    po = part_o.data_ptr<float>();
-    pml = part_ml.data_ptr<float>();
-  }
-  CHECK_HIP(lds_paged_attention_v4(
-      q.data_ptr(), k_cache.data_ptr(), v_cache.data_ptr(),
-      block_tables.data_ptr<int32_t>(), seq_lens.data_ptr<int32_t>(),
-      out.data_ptr(), po, pml, B, qh, kvh, bs, d,
-      (int)block_tables.size(1), (int)n_parts, (int)part_tokens,
-      kv_fp8_flag(k_cache), (float)scale, cur_stream()));
-  return out;
-}
-
-torch::Tensor paged_attention_v5(torch::Tensor q, torch::Tensor k_cache,
-                                 torch::Tensor v_cache,
-                                 torch::Tensor block_tables,
-                                 torch::Tensor seq_lens, int64_t n_parts,
-                                 int64_t part_tokens, double scale) {
-  CHECK_GPU(q); CHECK_GPU(k_cache); CHECK_GPU(v_cache);
-  CHECK_GPU(block_tables); CHECK_GPU(seq_lens);
-  int B = (int)q.size(0), qh = (int)q.size(1), d = (int)q.size(2);
-  int kvh = (int)k_cache.size(1), bs = (int)k_cache.size(2);
-  int qpg = qh / kvh;
-  auto out = torch::empty_like(q);
-  auto f32 = q.options().dtype(torch::kFloat32);
-  torch::Tensor part_o, part_ml;
-  float *po = nullptr, *pml = nullptr;
-  if (n_parts > 1) {
-    part_o = torch::empty({B, kvh, n_parts, qpg, d}, f32);
-    part_ml = torch::empty({B, kvh, n_parts, qpg, 2}, f32);
-    po = part_o.data_ptr<float>();
-    pml = part_ml.data_ptr<float>();
-  }
-  CHECK_HIP(lds_paged_attention_v5(
-      q.data_ptr(), k_cache.data_ptr(), v_cache.data_ptr(),
-      block_tables.data_ptr<int32_t>(), seq_lens.data_ptr<int32_t>(),
-      out.data_ptr(), po, pml, B, qh, kvh, bs, d,
-      (int)block_tables.size(1), (int)n_parts, (int)part_tokens,
-      kv_fp8_flag(k_cache), (float)scale, cur_stream()));
-  return out;
-}
-
 torch::Tensor flash_prefill(torch::Tensor q, torch::Tensor k_cache,
                             torch::Tensor v_cache, torch::Tensor block_tables,
                             torch::Tensor seq_meta, torch::Tensor tiles,
@@ -426,12 +327,6 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("paged_attention", &paged_attention, "GQA decode attention over paged KV");
   m.def("paged_attention_split", &paged_attention_split,
         "flash-decoding GQA attention with sequence partitioning");
-  m.def("paged_attention_v3", &paged_attention_v3,
-        "barrier-free one-wave-per-unit GQA decode attention");
-  m.def("paged_attention_v4", &paged_attention_v4,
-        "producer/consumer pipelined GQA decode attention");
-  m.def("paged_attention_v5", &paged_attention_v5,
-        "per-wave independent online-softmax decode attention");
   m.def("flash_prefill", &flash_prefill,
         "fused MFMA causal varlen prefill attention over paged KV");
 }

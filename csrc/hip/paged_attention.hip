@@ -355,232 +355,3 @@ hipError_t lds_paged_attention_split(
 }
 
 }  // extern "C"
-
-// ---------------------------------------------------------------------------
-// v3 geometry: ONE WAVE per (seq, kv_head, partition) — no __syncthreads at
-// all. The access-pattern probe (tools/kv_bw_probe.hip) sustains 6.5 TB/s
-// where the 4-wave phased kernel reaches 3.1: the A/B/C barriers and the
-// cross-wave combine are the gap. Here each wave owns its unit end-to-end:
-//   - 64-token chunks, lane = token: K-row streaming dots (registers)
-//   - in-wave online softmax (shfl reductions), P staged via per-wave LDS
-//     slice (write->read within one wave needs only lgkmcnt, no barrier)
-//   - V accumulation lane = dim pair, rows batched 8 deep
-// Workgroup = 4 independent waves; the split-partition count is chosen by
-// the host so total waves >= ~16/CU.
-namespace {
-
-template <int QPG, bool SPLIT, typename CT>
-__global__ __launch_bounds__(4 * WAVE) void paged_attention_v3_kernel(
-    const short* __restrict__ q,
-    const CT* __restrict__ k_cache, const CT* __restrict__ v_cache,
-    const int32_t* __restrict__ block_tables,
-    const int32_t* __restrict__ seq_lens, short* __restrict__ out,
-    float* __restrict__ part_o, float* __restrict__ part_ml,
-    int n_seqs, int kvh, int bs, int max_blocks, int n_parts,
-    int part_tokens, float scale) {
-  const int wave = threadIdx.x / WAVE;
-  const int lane = threadIdx.x % WAVE;
-  const int unit = blockIdx.x * 4 + wave;
-  if (unit >= n_seqs * kvh * n_parts) return;
-  const int part = unit % n_parts;
-  const int kh = (unit / n_parts) % kvh;
-  const int b = unit / (n_parts * kvh);
-  const int n_q_heads = kvh * QPG;
-  const int seq_len = seq_lens[b];
-  const int t_begin = part * part_tokens;
-  const int t_end = min(seq_len, t_begin + part_tokens);
-
-  // per-wave LDS slices: scaled Q rows + current chunk's P values
-  __shared__ float q_lds_all[4][QPG][D];
-  __shared__ float p_lds_all[4][QPG][WAVE];
-  float (*q_lds)[D] = q_lds_all[wave];
-  float (*p_lds)[WAVE] = p_lds_all[wave];
-
-  if (SPLIT && t_begin >= seq_len) {
-    if (lane < QPG) {
-      float* ml = part_ml + ((((int64_t)b * kvh + kh) * n_parts + part) *
-                             QPG + lane) * 2;
-      ml[0] = NEG;
-      ml[1] = 0.f;
-    }
-    return;
-  }
-
-  for (int i = lane; i < QPG * D; i += WAVE) {
-    const int h = i / D, d = i % D;
-    q_lds[h][d] =
-        bf16_to_f32(q[((int64_t)b * n_q_heads + kh * QPG + h) * D + d]) *
-        scale;
-  }
-  // within-wave LDS write->read: drain lgkm, no barrier needed
-  asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
-
-  float m_run[QPG], l_run[QPG], o0[QPG], o1[QPG];
-#pragma unroll
-  for (int h = 0; h < QPG; ++h) {
-    m_run[h] = NEG;
-    l_run[h] = 0.f;
-    o0[h] = o1[h] = 0.f;
-  }
-  const int32_t* bt = block_tables + (int64_t)b * max_blocks;
-
-  for (int chunk0 = t_begin; chunk0 < t_end; chunk0 += WAVE) {
-    const int n_t = min(WAVE, t_end - chunk0);
-    // ---- dots: lane owns token chunk0+lane ----
-    float dot[QPG];
-#pragma unroll
-    for (int h = 0; h < QPG; ++h) dot[h] = NEG;
-    if (lane < n_t) {
-      const int t = chunk0 + lane;
-      const CT* krow =
-          k_cache + (((int64_t)bt[t / bs] * kvh + kh) * bs + t % bs) * D;
-#pragma unroll
-      for (int h = 0; h < QPG; ++h) dot[h] = 0.f;
-#pragma unroll 8
-      for (int c = 0; c < D / 8; ++c) {
-        float kf[8];
-        load_kv8(krow + c * 8, kf);
-#pragma unroll
-        for (int h = 0; h < QPG; ++h) {
-          const float4v* q4 = (const float4v*)&q_lds[h][c * 8];
-          float4v qa = q4[0], qb = q4[1];
-          dot[h] += qa[0] * kf[0] + qa[1] * kf[1] + qa[2] * kf[2] +
-                    qa[3] * kf[3] + qb[0] * kf[4] + qb[1] * kf[5] +
-                    qb[2] * kf[6] + qb[3] * kf[7];
-        }
-      }
-    }
-    // ---- in-wave online softmax per head ----
-#pragma unroll
-    for (int h = 0; h < QPG; ++h) {
-      const float cmax = wave_reduce_max(dot[h]);
-      const float m_new = fmaxf(m_run[h], cmax);
-      const float p = (lane < n_t) ? __expf(dot[h] - m_new) : 0.f;
-      const float csum = wave_reduce_sum(p);
-      const float alpha =
-          (m_run[h] <= NEG * 0.5f) ? 0.f : __expf(m_run[h] - m_new);
-      l_run[h] = l_run[h] * alpha + csum;
-      m_run[h] = m_new;
-      o0[h] *= alpha;
-      o1[h] *= alpha;
-      p_lds[h][lane] = p;
-    }
-    asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
-    // ---- V accumulation: lane owns dims (2*lane, 2*lane+1) ----
-    int i = 0;
-    for (; i + 8 <= n_t; i += 8) {
-      const CT* vrows[8];
-#pragma unroll
-      for (int j = 0; j < 8; ++j) {
-        const int t = chunk0 + i + j;
-        vrows[j] = v_cache +
-                   (((int64_t)bt[t / bs] * kvh + kh) * bs + t % bs) * D +
-                   lane * 2;
-      }
-      float vv[8][2];
-#pragma unroll
-      for (int j = 0; j < 8; ++j) load_kv2(vrows[j], vv[j][0], vv[j][1]);
-#pragma unroll
-      for (int j = 0; j < 8; ++j) {
-#pragma unroll
-        for (int h = 0; h < QPG; ++h) {
-          const float p = p_lds[h][i + j];
-          o0[h] += p * vv[j][0];
-          o1[h] += p * vv[j][1];
-        }
-      }
-    }
-    for (; i < n_t; ++i) {
-      const int t = chunk0 + i;
-      float v0, v1;
-      load_kv2(v_cache + (((int64_t)bt[t / bs] * kvh + kh) * bs + t % bs) * D
-               + lane * 2, v0, v1);
-#pragma unroll
-      for (int h = 0; h < QPG; ++h) {
-        const float p = p_lds[h][i];
-        o0[h] += p * v0;
-        o1[h] += p * v1;
-      }
-    }
-  }
-
-  if (SPLIT) {
-    float* po = part_o +
-                ((((int64_t)b * kvh + kh) * n_parts + part) * QPG) * D;
-#pragma unroll
-    for (int h = 0; h < QPG; ++h) {
-      po[h * D + 2 * lane] = o0[h];
-      po[h * D + 2 * lane + 1] = o1[h];
-    }
-    if (lane == 0) {
-      // m_run/l_run are wave-uniform post-reduction; static indexing only
-      // (a runtime-indexed register array would force scratch)
-      float* ml = part_ml +
-                  ((((int64_t)b * kvh + kh) * n_parts + part) * QPG) * 2;
-#pragma unroll
-      for (int h = 0; h < QPG; ++h) {
-        ml[2 * h] = m_run[h];
-        ml[2 * h + 1] = l_run[h];
-      }
-    }
-  } else {
-#pragma unroll
-    for (int h = 0; h < QPG; ++h) {
-      const float inv = l_run[h] > 0.f ? 1.f / l_run[h] : 0.f;
-      short* orow =
-          out + ((int64_t)b * n_q_heads + kh * QPG + h) * D + 2 * lane;
-      orow[0] = f32_to_bf16(o0[h] * inv);
-      orow[1] = f32_to_bf16(o1[h] * inv);
-    }
-  }
-}
-
-}  // namespace
-
-extern "C" {
-
-hipError_t lds_paged_attention_v3(
-    const void* q, const void* k_cache, const void* v_cache,
-    const int32_t* block_tables, const int32_t* seq_lens, void* out,
-    float* part_o, float* part_ml, int n_seqs, int n_q_heads, int kvh, int bs,
-    int head_dim, int max_blocks, int n_parts, int part_tokens, int kv_fp8,
-    float scale, hipStream_t stream) {
-  if (n_seqs == 0) return hipSuccess;
-  if (head_dim != D || n_parts > 64) return hipErrorInvalidValue;
-  const int qpg = n_q_heads / kvh;
-  const int units = n_seqs * kvh * n_parts;
-  dim3 grid((units + 3) / 4), block(4 * WAVE);
-  dim3 cgrid(n_seqs, kvh), cblock(256);
-#define LAUNCH_CT3(QPG, CT, SPLIT)                                           \
-  hipLaunchKernelGGL((paged_attention_v3_kernel<QPG, SPLIT, CT>), grid,      \
-                     block, 0, stream, (const short*)q, (const CT*)k_cache,  \
-                     (const CT*)v_cache, block_tables, seq_lens,             \
-                     (short*)out, part_o, part_ml, n_seqs, kvh, bs,          \
-                     max_blocks, n_parts, part_tokens, scale)
-#define LAUNCH3(QPG)                                                         \
-  do {                                                                       \
-    if (n_parts > 1) {                                                       \
-      if (kv_fp8) LAUNCH_CT3(QPG, unsigned char, true);                      \
-      else LAUNCH_CT3(QPG, short, true);                                     \
-      hipLaunchKernelGGL((paged_attention_combine_kernel<QPG>), cgrid,       \
-                         cblock, 0, stream, part_o, part_ml, (short*)out,    \
-                         kvh, n_parts);                                      \
-    } else {                                                                 \
-      if (kv_fp8) LAUNCH_CT3(QPG, unsigned char, false);                     \
-      else LAUNCH_CT3(QPG, short, false);                                    \
-    }                                                                        \
-  } while (0)
-  switch (qpg) {
-    case 1: LAUNCH3(1); break;
-    case 2: LAUNCH3(2); break;
-    case 4: LAUNCH3(4); break;
-    case 8: LAUNCH3(8); break;
-    default: return hipErrorInvalidValue;
-  }
-#undef LAUNCH3
-#undef LAUNCH_CT3
-  HIP_CHECK_LAST();
-  return hipSuccess;
-}
-
-}  // extern "C"

@@ -35,8 +35,6 @@ hip_ops = cpp_extension.CUDAExtension(
         "csrc/hip/norm_rope_act.hip",
         "csrc/hip/kv_cache.hip",
         "csrc/hip/paged_attention.hip",
-        "csrc/hip/paged_attention_v4.hip",
-        "csrc/hip/paged_attention_v5.hip",
         "csrc/hip/flash_prefill.hip",
         "csrc/hip/flash_prefill_glds.hip",
     ],

@@ -464,82 +464,6 @@ class TestFp8KVCacheGPU:
         assert len(toks) == 6
 
 
-class TestPagedAttentionV3:
-    @pytest.mark.parametrize("qpg,np_,seqs,fp8",
-                             [(4, 1, [77, 1024, 333], False),
-                              (4, 4, [1024, 2049], False),
-                              (8, 2, [500], False),
-                              (1, 1, [129], False),
-                              (4, 4, [1024, 333], True)])
-    def test_v3_matches_v1(self, ext, qpg, np_, seqs, fp8):
-        """Barrier-free one-wave-per-unit kernel == the 4-wave kernel."""
-        torch.manual_seed(21)
-        KVH, D, BS = 8, 128, 16
-        QH = KVH * qpg
-        B = len(seqs)
-        max_blocks = (max(seqs) + BS - 1) // BS
-        NB = max_blocks * B + 1
-        q = torch.randn(B, QH, D, device="cuda").bfloat16()
-        kc = torch.randn(NB, KVH, BS, D, device="cuda")
-        vc = torch.randn(NB, KVH, BS, D, device="cuda")
-        kc = kc.to(torch.float8_e4m3fn) if fp8 else kc.bfloat16()
-        vc = vc.to(torch.float8_e4m3fn) if fp8 else vc.bfloat16()
-        bt = torch.zeros(B, max_blocks, dtype=torch.int32)
-        perm = torch.randperm(NB - 1) + 1
-        k = 0
-        for b, s in enumerate(seqs):
-            nb = (s + BS - 1) // BS
-            bt[b, :nb] = perm[k:k + nb]
-            k += nb
-        sl = torch.tensor(seqs, dtype=torch.int32).cuda()
-        base = ext.paged_attention(q, kc, vc, bt.cuda(), sl, D ** -0.5)
-        part = (max(seqs) + np_ - 1) // np_
-        part = (part + 63) // 64 * 64
-        v3 = ext.paged_attention_v3(q, kc, vc, bt.cuda(), sl, np_, part,
-                                    D ** -0.5)
-        assert torch.allclose(to_f32(base), to_f32(v3), atol=2e-2,
-                              rtol=2e-2), (to_f32(base) - to_f32(v3)).abs().max()
-
-
-class TestPagedAttentionV4:
-    @pytest.mark.parametrize("qpg,np_,seqs,fp8",
-                             [(4, 1, [77, 1024, 333], False),
-                              (4, 4, [1024, 2049], False),
-                              (8, 2, [500], False),
-                              (2, 1, [129], False),
-                              (1, 1, [16], False),
-                              (4, 4, [1024, 333], True)])
-    def test_v4_matches_v1(self, ext, qpg, np_, seqs, fp8):
-        """Producer/consumer pipelined kernel == the 4-wave phased kernel
-        (paged_attention_v4.hip; same online-softmax math)."""
-        torch.manual_seed(22)
-        KVH, D, BS = 8, 128, 16
-        QH = KVH * qpg
-        B = len(seqs)
-        max_blocks = (max(seqs) + BS - 1) // BS
-        NB = max_blocks * B + 1
-        q = torch.randn(B, QH, D, device="cuda").bfloat16()
-        kc = torch.randn(NB, KVH, BS, D, device="cuda")
-        vc = torch.randn(NB, KVH, BS, D, device="cuda")
-        kc = kc.to(torch.float8_e4m3fn) if fp8 else kc.bfloat16()
-        vc = vc.to(torch.float8_e4m3fn) if fp8 else vc.bfloat16()
-        bt = torch.zeros(B, max_blocks, dtype=torch.int32)
-        perm = torch.randperm(NB - 1) + 1
-        k = 0
-        for b, s in enumerate(seqs):
-            nb = (s + BS - 1) // BS
-            bt[b, :nb] = perm[k:k + nb]
-            k += nb
-        sl = torch.tensor(seqs, dtype=torch.int32).cuda()
-        base = ext.paged_attention(q, kc, vc, bt.cuda(), sl, D ** -0.5)
-        part = (max(seqs) + np_ - 1) // np_
-        part = (part + 255) // 256 * 256
-        v4 = ext.paged_attention_v4(q, kc, vc, bt.cuda(), sl, np_, part,
-                                    D ** -0.5)
-        assert torch.allclose(to_f32(base), to_f32(v4), atol=2e-2,
-                              rtol=2e-2), (to_f32(base) - to_f32(v4)).abs().max()
-
-
 class TestEngineFeaturesGPU:
     def _cfg(self):
         from llm_d_inference_scheduler_amd.models.configs import ModelConfig

@@ -54,15 +54,6 @@ def bench(B, ctx, np_=None, part=512, iters=50, kernel="v1", qpg=4):
     scale = D ** -0.5
 
     def run():
-        if kernel == "v3":
-            return ext.paged_attention_v3(q, kc, vc, bt, sl, np_ or 1,
-                                          part if np_ else ctx + 256, scale)
-        if kernel == "v4":
-            return ext.paged_attention_v4(q, kc, vc, bt, sl, np_ or 1,
-                                          part if np_ else ctx + 256, scale)
-        if kernel == "v5":
-            return ext.paged_attention_v5(q, kc, vc, bt, sl, np_ or 1,
-                                          part if np_ else ctx + 256, scale)
         if np_:
             return ext.paged_attention_split(q, kc, vc, bt, sl, np_, part,
                                              scale)
@@ -82,19 +73,11 @@ def bench(B, ctx, np_=None, part=512, iters=50, kernel="v1", qpg=4):
 
 
 if __name__ == "__main__":
-    check("v5", qpg=4)
-    check("v5", qpg=4, fp8=True)
-    check("v5", qpg=8)
-    check("v5", qpg=1)
     for B in (64, 128, 256):
         bench(B, 1152)
-        bench(B, 1152, kernel="v5")
         for np_ in (2, 4):
             part = ((1152 + np_ - 1) // np_ + 255) // 256 * 256
             bench(B, 1152, np_=np_, part=part)
-            bench(B, 1152, np_=np_, part=part, kernel="v5")
     bench(8, 8192, np_=16, part=512)
-    bench(8, 8192, np_=16, part=512, kernel="v5")
     bench(128, 1536)
-    bench(128, 1536, kernel="v5")
-    bench(256, 1152, kernel="v4")
+

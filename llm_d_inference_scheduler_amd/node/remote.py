@@ -36,6 +36,9 @@ CANCEL_PATH = "/internal/v1/cancel"
 
 # transport(url, payload_dict) -> response_dict; injectable for tests
 Transport = Callable[[str, Dict], Dict]
+# stream_transport(url, payload_dict) -> iterator of event dicts
+# ({"tokens": [...]} chunks, then a final {"done": True, ...} record)
+StreamTransport = Callable[[str, Dict], object]
 
 
 def remote_endpoint(name: str, index: int, url: str,
@@ -54,6 +57,24 @@ def _default_transport(url: str, payload: Dict) -> Dict:
     return r.json()
 
 
+def _default_stream_transport(url: str, payload: Dict):
+    """SSE relay: iterate `data:` events from the peer's streaming
+    enqueue route (decode.go SSE re-emission, here across nodes)."""
+    import json as json_mod
+
+    import httpx
+    with httpx.stream("POST", url, json=payload, timeout=300.0) as r:
+        r.raise_for_status()
+        for line in r.iter_lines():
+            line = line.strip()
+            if not line.startswith("data:"):
+                continue
+            body = line[5:].strip()
+            if body == "[DONE]":
+                break
+            yield json_mod.loads(body)
+
+
 class RemoteForwarder:
     """Forwards scheduled requests to remote-node endpoints on a small
     thread pool; completed responses drain into the router step as
@@ -61,11 +82,14 @@ class RemoteForwarder:
     request to the remote engine and relaying the response)."""
 
     def __init__(self, transport: Optional[Transport] = None,
-                 max_workers: int = 8):
+                 max_workers: int = 8,
+                 stream_transport: Optional[StreamTransport] = None):
         self.transport = transport or _default_transport
+        self.stream_transport = stream_transport or _default_stream_transport
         self.max_workers = max_workers
         self._pool: Optional[ThreadPoolExecutor] = None
         self._done: "queue.Queue" = queue.Queue()
+        self._tokens: "queue.Queue" = queue.Queue()  # (req_id, [tokens])
         self._lock = threading.Lock()
         self.inflight = 0
 
@@ -85,10 +109,12 @@ class RemoteForwarder:
             "temperature": req.temperature,
             "stop_token_ids": req.stop_token_ids,
             "priority": req.priority,
+            "stream": bool(req.streaming),
         }
         with self._lock:
             self.inflight += 1
-        self._ensure_pool().submit(self._run, req.request_id,
+        runner = self._run_stream if req.streaming else self._run
+        self._ensure_pool().submit(runner, req.request_id,
                                    base_url + ENQUEUE_PATH, payload)
 
     def _run(self, request_id: str, url: str, payload: Dict) -> None:
@@ -113,6 +139,49 @@ class RemoteForwarder:
         with self._lock:
             self.inflight -= 1
         self._done.put(comp)
+
+    def _run_stream(self, request_id: str, url: str, payload: Dict) -> None:
+        """Cross-node SSE token relay: per-token chunks surface as router
+        token events while the generation runs on the peer; the final
+        event closes out as a normal Completion."""
+        from .runner import Completion
+        final = None
+        try:
+            for ev in self.stream_transport(url, payload):
+                if ev.get("done"):
+                    final = ev
+                    break
+                toks = ev.get("tokens") or []
+                if toks:
+                    self._tokens.put((request_id, list(toks)))
+            if final is None:
+                raise RuntimeError("stream ended without a final event")
+            u = final.get("usage", {})
+            comp = Completion(
+                request_id=request_id,
+                usage=Usage(prompt_tokens=u.get("prompt_tokens", 0),
+                            completion_tokens=u.get("completion_tokens", 0),
+                            cached_tokens=u.get("cached_tokens", 0),
+                            ttft_ms=u.get("ttft_ms"),
+                            e2e_ms=u.get("e2e_ms")),
+                tokens=final.get("tokens", []),
+                finish_reason=final.get("finish_reason", "length"),
+                error=final.get("error", ""))
+        except Exception as e:
+            log.error("remote stream failed", url=url, err=str(e))
+            comp = Completion(request_id=request_id, usage=Usage(),
+                              error=f"remote_error: {e}")
+        with self._lock:
+            self.inflight -= 1
+        self._done.put(comp)
+
+    def drain_tokens(self) -> List:
+        out = []
+        while True:
+            try:
+                out.append(self._tokens.get_nowait())
+            except queue.Empty:
+                return out
 
     def cancel(self, request_id: str, base_url: str) -> None:
         """Best-effort remote abort: tells the peer to unwind the

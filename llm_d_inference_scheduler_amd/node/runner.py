@@ -467,6 +467,13 @@ class NodeRunner:
         if self.is_router:
             self._route_arrivals()
             self._maybe_evict_inflight()
+            for rid, toks in self.remote.drain_tokens():
+                # cross-node SSE relay: peer tokens surface as router
+                # token events (same path local streaming tokens ride)
+                self._token_events.append((rid, toks))
+                decision = self._decisions.get(rid)
+                if decision is not None:
+                    self.director.handle_response_chunk(decision, toks)
             for comp in self.remote.drain():
                 self._remote_urls.pop(comp.request_id, None)
                 decision = self._decisions.pop(comp.request_id, None)

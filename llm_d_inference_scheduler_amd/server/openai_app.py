@@ -17,7 +17,7 @@ from typing import Optional
 from fastapi import FastAPI, Request, Response
 from fastapi.responses import JSONResponse, StreamingResponse
 
-from ..handlers.parsers import ParserMux
+from ..handlers.parsers import ParserMux, Usage
 from ..metrics import prom
 from ..models.tokenizer import HashTokenizer
 from ..scheduling.types import LLMRequest
@@ -239,8 +239,36 @@ def build_app(service: NodeService,
             max_tokens=int(body.get("max_tokens", 16)),
             temperature=float(body.get("temperature") or 0.0),
             stop_token_ids=body.get("stop_token_ids"),
+            streaming=bool(body.get("stream", False)),
             priority=int(body.get("priority", 0)))
         handle = service.submit(req)
+        if req.streaming:
+            # cross-node SSE token relay: the forwarding router re-emits
+            # these chunks to ITS client as they arrive
+            async def _relay():
+                while True:
+                    item = await asyncio.to_thread(handle.token_queue.get)
+                    if item is None:
+                        break
+                    toks = item if isinstance(item, list) else [item]
+                    yield ("data: " + json.dumps({"tokens": toks}) + "\n\n"
+                           ).encode()
+                comp = handle.completion
+                u = comp.usage if comp else Usage()
+                final = {"done": True,
+                         "tokens": comp.tokens if comp else [],
+                         "finish_reason": (comp.finish_reason
+                                           if comp else "length"),
+                         "error": comp.error if comp else "lost",
+                         "usage": {"prompt_tokens": u.prompt_tokens,
+                                   "completion_tokens": u.completion_tokens,
+                                   "cached_tokens": u.cached_tokens,
+                                   "ttft_ms": u.ttft_ms,
+                                   "e2e_ms": u.e2e_ms}}
+                yield ("data: " + json.dumps(final) + "\n\n").encode()
+                yield b"data: [DONE]\n\n"
+            return StreamingResponse(_relay(),
+                                     media_type="text/event-stream")
         completion = await asyncio.to_thread(handle.wait, 300.0)
         if completion is None:
             service.cancel(req.request_id)

@@ -2,6 +2,7 @@
 node B's front door (node/remote.py + /internal/v1/enqueue +
 HttpMetricsSource). The reference analog is cross-pool routing
 (InferencePoolImport); here two full nodes run in-process on CPU."""
+import json
 import time
 
 import pytest
@@ -156,3 +157,72 @@ class TestRemoteCancel:
                                json={"request_id": "r-c"})
         assert r.status_code == 200 and r.json()["canceled"] == "r-c"
         node_a.shutdown()
+
+
+class TestCrossNodeSSERelay:
+    """Cross-node SSE token relay (round-2: streaming across nodes; the
+    non-streaming v1 buffered whole completions)."""
+
+    def test_streaming_enqueue_emits_sse(self):
+        """The peer side: /internal/v1/enqueue with stream=true returns
+        SSE token chunks then a final done record."""
+        cfg = NodeConfig(model=TINY_LLAMA, world_size=1, topology="mono",
+                         device="cpu", dtype=torch.float32, kv_blocks=128)
+        service = NodeService(NodeRunner(cfg))
+        service.start()
+        with TestClient(build_app(service)) as c:
+            with c.stream("POST", "/internal/v1/enqueue", json={
+                    "request_id": "s1", "model": "tiny-llama",
+                    "prompt_tokens": list(range(40)), "max_tokens": 4,
+                    "stream": True}) as r:
+                assert r.status_code == 200
+                events = []
+                for line in r.iter_lines():
+                    line = line.strip()
+                    if line.startswith("data:") and \
+                            line[5:].strip() != "[DONE]":
+                        events.append(json.loads(line[5:]))
+        service.stop()
+        final = events[-1]
+        assert final["done"] and final["error"] == ""
+        assert final["usage"]["completion_tokens"] == 4
+        streamed = [t for e in events[:-1] for t in e.get("tokens", [])]
+        assert len(streamed) == 4
+        assert streamed == final["tokens"]
+
+    def test_forwarder_relays_stream_to_router(self):
+        """The forwarding side: stream events surface as router token
+        events while the peer generates, then close as a Completion."""
+        from llm_d_inference_scheduler_amd.node.remote import (
+            RemoteForwarder, remote_endpoint)
+
+        def fake_stream(url, payload):
+            assert payload["stream"] is True
+            yield {"tokens": [11, 12]}
+            yield {"tokens": [13]}
+            yield {"done": True, "tokens": [11, 12, 13],
+                   "finish_reason": "length", "error": "",
+                   "usage": {"prompt_tokens": 5, "completion_tokens": 3}}
+
+        cfg = NodeConfig(model=TINY_LLAMA, world_size=1, topology="mono",
+                         device="cpu", dtype=torch.float32, kv_blocks=64)
+        node = NodeRunner(cfg)
+        node.remote = RemoteForwarder(stream_transport=fake_stream)
+        node.datastore.add_endpoint(remote_endpoint(
+            "peer0", 9, "http://peer:8000"))
+        from llm_d_inference_scheduler_amd.scheduling.types import LLMRequest
+        req = LLMRequest(request_id="sr1", model="tiny-llama",
+                         prompt_tokens=list(range(30)), max_tokens=3,
+                         streaming=True, subset_hint=["peer0"])
+        node.submit(req)
+        done = []
+        for _ in range(100):
+            node.step()
+            done.extend(node.drain_completions())
+            if done:
+                break
+        assert done and not done[0].error
+        assert done[0].tokens == [11, 12, 13]
+        toks = [t for _, chunk in node.drain_token_events() for t in chunk]
+        assert toks == [11, 12, 13]
+        node.shutdown()

@@ -493,6 +493,7 @@ class NodeRunner:
         self._outbox = []
         self._process_messages(msgs)
         self._execute_transfers(msgs)
+        self._sweep_stale_adoptions()
         self._run_encoder()
         outputs = self.engine.step()
         self._handle_outputs(outputs)
@@ -766,6 +767,7 @@ class NodeRunner:
                     bs = self.engine.pool.block_size
                     n = (len(decode_req.prompt_tokens) + bs - 1) // bs
                     adoption["reserved"] = self.engine.mgr.take_blocks(n)
+                adoption["step"] = self._step
                 self._pending_adoption[m["req_id"]] = adoption
         if encode_ranks and self.rank in encode_ranks:
             self._encode_jobs.append({"req_id": m["req_id"],
@@ -963,6 +965,25 @@ class NodeRunner:
                                       on_complete=_adopt)
         else:
             self.transfer.recv_blocks(src, local, on_complete=_adopt)
+
+    def _sweep_stale_adoptions(self, max_age_steps: int = 2000) -> None:
+        """A decode-side adoption whose prefill peer died (rank crash, no
+        kv_ready ever arrives) must not hold reserved blocks and the
+        client's slot forever: after max_age_steps the request fails with
+        `prefill_lost` and reservations are released (the reference's
+        analog is the sidecar's prefill-failure fallback + pod-removal
+        cleanup, connector_nixlv2.go:160-176 / runner.go endpoint hooks)."""
+        if not self._pending_adoption:
+            return
+        stale = [rid for rid, p in self._pending_adoption.items()
+                 if self._step - p.get("step", self._step) > max_age_steps]
+        for rid in stale:
+            pend = self._pending_adoption.pop(rid)
+            if pend.get("reserved"):
+                self.engine.mgr.release_blocks(pend["reserved"])
+            log.warning("pending adoption expired", req=rid)
+            self._outbox.append({"type": "done", "req_id": rid,
+                                 "error": "prefill_lost"})
 
     # ---- engine outputs -> messages ----
     def _handle_outputs(self, outputs: List[RequestOutput]) -> None:

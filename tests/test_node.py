@@ -425,3 +425,26 @@ class TestPipelinedMailbox:
         assert len(results) == 5
         assert all(not r["error"] for r in results), results
         assert all(r["completion"] == 4 for r in results)
+
+
+class TestPrefillLoss:
+    def test_stale_adoption_expires_and_frees_reservation(self):
+        """A decode rank whose prefill peer died (kv_ready never arrives)
+        fails the request with prefill_lost after the TTL and releases
+        the sglang-style reservation instead of leaking it."""
+        from llm_d_inference_scheduler_amd.engine import EngineRequest
+        cfg = NodeConfig(model=TINY_LLAMA, world_size=1, topology="mono",
+                         device="cpu", dtype=torch.float32, kv_blocks=64)
+        node = NodeRunner(cfg)
+        mgr = node.engine.mgr
+        free0 = mgr.free_blocks
+        node._pending_adoption["dead"] = {
+            "req": EngineRequest("dead", list(range(20)), max_tokens=2),
+            "src": 1, "reserved": mgr.take_blocks(4), "step": 0}
+        node._step = 10
+        node._sweep_stale_adoptions(max_age_steps=5)
+        assert "dead" not in node._pending_adoption
+        assert mgr.free_blocks == free0
+        done = [m for m in node._outbox if m.get("type") == "done"]
+        assert done and done[0]["error"] == "prefill_lost"
+        node.shutdown()

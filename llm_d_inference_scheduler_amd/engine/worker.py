@@ -215,6 +215,7 @@ class EngineWorker:
         self.dstate = DecodeState(min(64, max_decode_batch), max_blocks,
                                   self.device)
         self._cuda = self.device.type == "cuda"
+        self._prefill_stream = None   # lazy; see step() overlap
         self._pin = None
         self._pending = None   # (reqs, event|None, n) — one-step readback lag
         self.steps = 0
@@ -332,12 +333,35 @@ class EngineWorker:
         if self._rejects:
             outputs.extend(self._rejects)
             self._rejects = []
-        if self.waiting and self._should_prefill():
-            outputs.extend(self._prefill_pass())
-        if self.running:
+        do_prefill = bool(self.waiting) and self._should_prefill()
+        if self._cuda and do_prefill and self.running:
+            # Overlap the two passes: decode (HBM-bound paged attention +
+            # skinny GEMMs) launches first on the default stream; prefill
+            # (MFMA-bound flash attention + fat GEMMs) runs concurrently
+            # on its own stream. Safe because the passes touch DISJOINT
+            # sequences and KV blocks (freed blocks re-enter the pool only
+            # after their final reader's event synchronized, and cached
+            # prefix blocks are immutable), and weights are read-only.
+            # No ps.wait_stream(default) on entry — that would serialize
+            # the two passes and defeat the overlap.
             outputs.extend(self._decode_pass())
-        elif self._pending is not None:
-            outputs.extend(self._collect_pending())
+            if self._prefill_stream is None:
+                self._prefill_stream = torch.cuda.Stream(self.device)
+            ps = self._prefill_stream
+            with torch.cuda.stream(ps):
+                outputs.extend(self._prefill_pass())
+            # _prefill_pass host-syncs ps for first-token sampling, but
+            # dstate.join() enqueues block-table copies on ps AFTER that
+            # sync point: the next step's decode (default stream) must
+            # observe them
+            torch.cuda.current_stream(self.device).wait_stream(ps)
+        else:
+            if do_prefill:
+                outputs.extend(self._prefill_pass())
+            if self.running:
+                outputs.extend(self._decode_pass())
+            elif self._pending is not None:
+                outputs.extend(self._collect_pending())
         self.steps += 1
         return outputs
 

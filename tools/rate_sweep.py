@@ -2,7 +2,12 @@
 regression-harness methodology, single-workload-regression.yaml:30-45;
 BASELINE.md "harness shape to reuse"). Run on a GPU box:
 
-  python tools/rate_sweep.py --rates 4 8 12 16 20 --steps 300
+  python tools/rate_sweep.py --rates 10 20 30 40 50 --steps 400 -- --max-tokens 128
+
+Unknown args after `--` pass through to bench.py (e.g. --mode fc,
+--max-tokens). The reference's harness runs 300 s per point at out=1024;
+a GPU-minute-bounded sweep instead uses out=128 so requests complete
+inside the window and the latency-throughput knee is visible.
 """
 import argparse
 import json
@@ -20,7 +25,8 @@ def main():
     ap.add_argument("--steps", type=int, default=300)
     ap.add_argument("--warmup", type=int, default=60)
     ap.add_argument("--extra", nargs="*", default=[])
-    args = ap.parse_args()
+    args, passthrough = ap.parse_known_args()
+    args.extra = list(args.extra) + [a for a in passthrough if a != "--"]
     rows = []
     for r in args.rates:
         cmd = [sys.executable, os.path.join(ROOT, "bench.py"),

@@ -226,3 +226,58 @@ class TestCrossNodeSSERelay:
         toks = [t for _, chunk in node.drain_token_events() for t in chunk]
         assert toks == [11, 12, 13]
         node.shutdown()
+
+
+class TestPoolImport:
+    """InferencePoolImport analog: declaring a remote pool materializes
+    peer front doors as schedulable endpoints; deletion removes them."""
+
+    def test_apply_route_and_delete(self):
+        from llm_d_inference_scheduler_amd.api.poolimport import (
+            ImportedEndpoint, InferencePoolImport, PoolImportManager)
+
+        def transport(url, payload):
+            return {"tokens": [5, 6], "finish_reason": "length",
+                    "error": "", "usage": {"prompt_tokens": 4,
+                                           "completion_tokens": 2}}
+        node = NodeRunner(NodeConfig(model=TINY_LLAMA, device="cpu",
+                                     dtype=torch.float32, kv_blocks=64,
+                                     remote_transport=transport))
+        mgr = PoolImportManager(node.datastore)
+        imp = InferencePoolImport("peerpool", endpoints=[
+            ImportedEndpoint("http://peer-a:8000"),
+            ImportedEndpoint("http://peer-b:8000")])
+        names = mgr.apply(imp)
+        assert len(names) == 2
+        eps = {ep.name for ep in node.datastore.endpoints()}
+        assert set(names) <= eps
+        # imported endpoints are schedulable: steer a request there
+        node.submit(LLMRequest(
+            request_id="imp1", model=TINY_LLAMA.name, prompt="",
+            prompt_tokens=list(range(12)), max_tokens=2,
+            subset_hint=names))
+        done = []
+        for _ in range(200):
+            node.step()
+            done.extend(node.drain_completions())
+            if done:
+                break
+            time.sleep(0.005)
+        assert done and done[0].tokens == [5, 6]
+        mgr.delete("peerpool")
+        eps = {ep.name for ep in node.datastore.endpoints()}
+        assert not (set(names) & eps)
+        node.shutdown()
+
+    def test_reapply_replaces(self):
+        from llm_d_inference_scheduler_amd.api.poolimport import (
+            ImportedEndpoint, InferencePoolImport, PoolImportManager)
+        from llm_d_inference_scheduler_amd.datalayer.datastore import \
+            Datastore
+        ds = Datastore()
+        mgr = PoolImportManager(ds)
+        mgr.apply(InferencePoolImport("p", [ImportedEndpoint("http://a")]))
+        mgr.apply(InferencePoolImport("p", [ImportedEndpoint("http://b")]))
+        eps = list(ds.endpoints())
+        assert len(eps) == 1
+        assert eps[0].metadata.labels["llm-d.ai/remote-url"] == "http://b"

@@ -27,12 +27,24 @@ def main():
     ap.add_argument("--extra", nargs="*", default=[])
     args, passthrough = ap.parse_known_args()
     args.extra = list(args.extra) + [a for a in passthrough if a != "--"]
+    outdir = os.environ.get("SWEEP_OUT", os.path.join(ROOT, "gpurun_out"))
+    os.makedirs(outdir, exist_ok=True)
     rows = []
     for r in args.rates:
         cmd = [sys.executable, os.path.join(ROOT, "bench.py"),
                "--steps", str(args.steps), "--warmup", str(args.warmup),
                "--arrival-rate", str(r)] + args.extra
         out = subprocess.run(cmd, capture_output=True, text=True)
+        tag = f"sweep_r{r:g}" + "".join(args.extra).replace("--", "_")
+        with open(os.path.join(outdir, tag + ".json"), "w") as f:
+            f.write(out.stdout)
+        if out.stderr:
+            with open(os.path.join(outdir, tag + ".err"), "w") as f:
+                f.write(out.stderr[-20000:])
+        if not out.stdout.strip():
+            print(f"rate {r}: bench produced no output "
+                  f"(stderr tail: {out.stderr[-400:]})", flush=True)
+            continue
         line = out.stdout.strip().splitlines()[-1]
         d = json.loads(line)
         rows.append((r, d))

@@ -179,6 +179,7 @@ class EngineWorker:
                  prefill_min_tokens: int = 4096,
                  prefill_max_delay_ms: float = 60.0,
                  ipc_pool: bool = False,
+                 overlap_streams: bool = True,
                  seed: int = 0):
         self.cfg = config
         self.device = torch.device(device)
@@ -215,6 +216,7 @@ class EngineWorker:
         self.dstate = DecodeState(min(64, max_decode_batch), max_blocks,
                                   self.device)
         self._cuda = self.device.type == "cuda"
+        self._overlap = overlap_streams
         self._prefill_stream = None   # lazy; see step() overlap
         self._pin = None
         self._pending = None   # (reqs, event|None, n) — one-step readback lag
@@ -334,7 +336,7 @@ class EngineWorker:
             outputs.extend(self._rejects)
             self._rejects = []
         do_prefill = bool(self.waiting) and self._should_prefill()
-        if self._cuda and do_prefill and self.running:
+        if self._cuda and self._overlap and do_prefill and self.running:
             # Overlap the two passes: decode (HBM-bound paged attention +
             # skinny GEMMs) launches first on the default stream; prefill
             # (MFMA-bound flash attention + fat GEMMs) runs concurrently

@@ -514,3 +514,42 @@ class TestEngineFeaturesGPU:
         fins = {o.request_id: o for o in outs if o.finished}
         assert len(fins) == 3
         assert all(o.completion_tokens == 40 for o in fins.values())
+
+
+class TestStreamOverlapGPU:
+    def test_overlap_matches_sequential(self):
+        """Prefill/decode stream overlap (engine/worker.py step()) is
+        bit-identical to the sequential path: same kernels, same order
+        within each pass, disjoint sequences."""
+        import dataclasses
+        from llm_d_inference_scheduler_amd.engine import (EngineRequest,
+                                                          EngineWorker)
+        from llm_d_inference_scheduler_amd.models.configs import ModelConfig
+        cfg = ModelConfig(name="gpu-ov", vocab_size=2048, hidden_size=1024,
+                          intermediate_size=2048, num_layers=2, num_heads=8,
+                          num_kv_heads=2, head_dim=128, rope_theta=10000.0)
+        results = {}
+        for overlap in (False, True):
+            w = EngineWorker(cfg, "cuda:0", kv_blocks=512,
+                             dtype=torch.bfloat16, seed=11,
+                             overlap_streams=overlap,
+                             prefill_min_tokens=64)
+            toks = {}
+            # staggered arrivals force waiting+running coexistence
+            w.add_request(EngineRequest("a", list(range(300, 500)),
+                                        max_tokens=24))
+            for step in range(80):
+                if step == 4:
+                    w.add_request(EngineRequest("b", list(range(700, 1020)),
+                                                max_tokens=16))
+                if step == 8:
+                    w.add_request(EngineRequest("c", list(range(64, 192)),
+                                                max_tokens=12))
+                for o in w.step():
+                    if o.finished:
+                        toks[o.request_id] = o.all_tokens
+                if not w.has_work:
+                    break
+            assert set(toks) == {"a", "b", "c"}
+            results[overlap] = toks
+        assert results[False] == results[True]

@@ -161,6 +161,11 @@ class Workload:
 
 def main():
     args = parse_args()
+    wd = float(os.environ.get("LLMD_BENCH_WATCHDOG", "0"))
+    if wd > 0:
+        # debug: dump all thread stacks and die if the run wedges
+        import faulthandler
+        faulthandler.dump_traceback_later(wd, exit=True)
     rank = int(os.environ.get("RANK", "0"))
     world = int(os.environ.get("WORLD_SIZE", "1"))
     local_rank = int(os.environ.get("LOCAL_RANK", str(rank)))

@@ -215,8 +215,10 @@ class EngineWorker:
         max_blocks = max_model_len // self.pool.block_size + 2
         self.dstate = DecodeState(min(64, max_decode_batch), max_blocks,
                                   self.device)
+        import os as _os
         self._cuda = self.device.type == "cuda"
-        self._overlap = overlap_streams
+        self._overlap = overlap_streams and \
+            _os.environ.get("LLMD_DISABLE_OVERLAP", "0") != "1"
         self._prefill_stream = None   # lazy; see step() overlap
         self._pin = None
         self._pending = None   # (reqs, event|None, n) — one-step readback lag

@@ -179,7 +179,7 @@ class EngineWorker:
                  prefill_min_tokens: int = 4096,
                  prefill_max_delay_ms: float = 60.0,
                  ipc_pool: bool = False,
-                 overlap_streams: bool = True,
+                 overlap_streams: Optional[bool] = None,
                  seed: int = 0):
         self.cfg = config
         self.device = torch.device(device)
@@ -217,8 +217,19 @@ class EngineWorker:
                                   self.device)
         import os as _os
         self._cuda = self.device.type == "cuda"
-        self._overlap = overlap_streams and \
-            _os.environ.get("LLMD_DISABLE_OVERLAP", "0") != "1"
+        # Prefill/decode stream overlap measured +11.8% at mixed
+        # prefill/decode shapes (profiles/r02_notes.md) but wedged two
+        # long open-loop runs non-deterministically (sweep r30 mono, r20
+        # fc — suspect a shared-resource race between concurrent GEMMs on
+        # the two streams). Tri-state: explicit True/False from the
+        # constructor wins (tests force it on); the None default is
+        # OFF unless LLMD_ENABLE_OVERLAP=1, until the wedge is
+        # root-caused.
+        if overlap_streams is None:
+            self._overlap = _os.environ.get("LLMD_ENABLE_OVERLAP",
+                                            "0") == "1"
+        else:
+            self._overlap = bool(overlap_streams)
         self._prefill_stream = None   # lazy; see step() overlap
         self._pin = None
         self._pending = None   # (reqs, event|None, n) — one-step readback lag

@@ -88,6 +88,9 @@ class NodeConfig:
     flow_control: bool = False
     fc_bands: List[BandConfig] = field(default_factory=list)
     fc_global_max_items: Optional[int] = None
+    # engine prefill/decode stream overlap (None = env default; see
+    # engine/worker.py)
+    overlap_streams: Any = None
     # >=2 so JSQ-bytes shard distribution is live by default
     # (controller.go:94-150; round-1 verdict flagged the single-shard
     # default as making it vacuous)
@@ -173,6 +176,7 @@ class NodeRunner:
             max_decode_batch=cfg.max_decode_batch,
             ttft_slo_ms=cfg.ttft_slo_ms,
             kv_cache_dtype=cfg.kv_cache_dtype, ipc_pool=want_ipc,
+            overlap_streams=cfg.overlap_streams,
             seed=cfg.seed)
         # encode role: vision tower + URL-deduped embedding cache
         self.encoder = None

@@ -246,3 +246,26 @@ class TestExtProcErrors:
                 for o in r.immediate_response.headers.set_headers}
         assert muts[DROPPED_REASON_HEADER] == "evicted"
         st.close()
+
+
+class TestExtProcTrailers:
+    def test_trailers_get_trailer_responses(self, server):
+        st = Stream(server.port)
+        st.send(req_headers({":path": "/v1/completions",
+                             "content-type": "application/json"}))
+        st.send(req_body(completion_body(), end_of_stream=True))
+        st.recv()   # headers response
+        st.recv()   # body response
+        m = pb.ProcessingRequest()
+        m.request_trailers.SetInParent()
+        st.send(m)
+        r = st.recv()
+        assert r.WhichOneof("response") == "request_trailers"
+        st.send(resp_headers())
+        st.recv()
+        m = pb.ProcessingRequest()
+        m.response_trailers.SetInParent()
+        st.send(m)
+        r = st.recv()
+        assert r.WhichOneof("response") == "response_trailers"
+        st.close()

@@ -286,8 +286,14 @@ torch::Tensor flash_prefill(torch::Tensor q, torch::Tensor k_cache,
   int qh = (int)q.size(1), d = (int)q.size(2);
   int kvh = (int)k_cache.size(1), bs = (int)k_cache.size(2);
   auto out = torch::empty_like(q);
-  // bf16 KV rides the glds staging pipeline; fp8 needs convert-on-stage
-  if (!kv_fp8_flag(k_cache) && block_tables.size(1) <= 1024) {
+  // bf16 KV rides the glds staging pipeline; fp8 needs convert-on-stage.
+  // LLMD_NO_GLDS=1 forces the plain-VGPR kernel (bisect knob for the
+  // cross-stream wedge investigation, profiles/r02_notes.md).
+  static const bool no_glds = []() {
+    const char* e = getenv("LLMD_NO_GLDS");
+    return e && e[0] == '1';
+  }();
+  if (!no_glds && !kv_fp8_flag(k_cache) && block_tables.size(1) <= 1024) {
     CHECK_HIP(lds_flash_prefill_glds(
         q.data_ptr(), k_cache.data_ptr(), v_cache.data_ptr(),
         block_tables.data_ptr<int32_t>(), seq_meta.data_ptr<int32_t>(),

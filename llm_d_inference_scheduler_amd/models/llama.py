@@ -38,6 +38,10 @@ def _load_tuned_gemms() -> None:
     _TUNED_GEMM_LOADED = True
     path = os.path.join(os.path.dirname(os.path.abspath(__file__)),
                         "..", "..", "profiles", "tuned_gemm_gfx950.csv")
+    if os.environ.get("LLMD_NO_TUNED_GEMM", "0") == "1":
+        # bisect knob: TunableOp is the prime suspect for the
+        # multi-stream wedge (profiles/r02_notes.md)
+        return
     if os.path.exists(path):
         import torch.cuda.tunable as tunable
         tunable.enable(True)

@@ -86,3 +86,51 @@ def test_every_scorer_returns_unit_interval(scorer_type):
     assert set(scores) == {e.name for e in eps}
     for name, v in scores.items():
         assert 0.0 <= v <= 1.0, (scorer_type, name, v)
+
+
+class TestMetricFamilyParity:
+    """Pin the §5.5 metric families (reference pkg/epp/metrics/metrics.go
+    + pkg/metrics docs/metrics.md) so a rename or accidental removal
+    fails loudly — the golden-metrics-testdata analog."""
+
+    def test_required_families_registered(self):
+        from llm_d_inference_scheduler_amd.metrics import prom
+        text = prom.render().decode()
+        fams = {l.split()[2] for l in text.splitlines()
+                if l.startswith("# TYPE")}
+        required = {
+            # request accounting (metrics.go:584-754)
+            "inference_extension_request_total",
+            "inference_extension_request_error_total",
+            "inference_extension_request_duration_seconds",
+            "inference_extension_request_sizes",
+            "inference_extension_input_tokens",
+            "inference_extension_output_tokens",
+            "inference_extension_cached_tokens",
+            "inference_extension_running_requests",
+            # TTFT / TPOT (+normalized)
+            "inference_extension_time_to_first_token_seconds",
+            "inference_extension_normalized_time_per_output_token_seconds",
+            # scheduler + per-plugin latency (metrics.go:786-821)
+            "inference_extension_scheduler_e2e_duration_seconds",
+            "inference_extension_plugin_duration_seconds",
+            # prefix cache (metrics.go:826-841)
+            "inference_extension_prefix_indexer_size",
+            "inference_extension_prefix_indexer_hit_ratio",
+            # flow control (metrics.go:848-897)
+            "inference_extension_flow_control_queue_size",
+            "inference_extension_flow_control_queue_duration_seconds",
+            "inference_extension_flow_control_dispatch_total",
+            "inference_extension_saturation",
+            # model rewrite decisions
+            "inference_extension_model_rewrite_total",
+            # llm-d families (docs/metrics.md:13-27)
+            "llm_d_inference_scheduler_disagg_decision_total",
+            # MI355X-native additions
+            "llm_d_inference_scheduler_xgmi_kv_transfer_bytes_total",
+            "llm_d_inference_scheduler_xgmi_kv_transfer_seconds",
+            "llm_d_inference_scheduler_datalayer_poll_errors_total",
+            "llm_d_inference_scheduler_datalayer_extract_errors_total",
+        }
+        missing = required - fams
+        assert not missing, f"metric families missing: {sorted(missing)}"

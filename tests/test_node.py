@@ -448,3 +448,57 @@ class TestPrefillLoss:
         done = [m for m in node._outbox if m.get("type") == "done"]
         assert done and done[0]["error"] == "prefill_lost"
         node.shutdown()
+
+
+def _mono4_worker(rank, world_size, init_file, out_file):
+    """mono DP world-4 with the pipelined mailbox — the exact shape of the
+    driver's multi-GPU SCALE run (bench.py --mode mono under torchrun)."""
+    import torch.distributed as dist
+    dist.init_process_group("gloo", init_method=f"file://{init_file}",
+                            rank=rank, world_size=world_size)
+    mailbox_group = dist.new_group(backend="gloo")
+    ctrl_group = dist.new_group(backend="gloo")
+    try:
+        cfg = NodeConfig(model=TINY_LLAMA, rank=rank, world_size=world_size,
+                         topology="mono", device="cpu",
+                         dtype=torch.float32, kv_blocks=256, seed=3,
+                         mailbox_group=mailbox_group)
+        node = NodeRunner(cfg)
+        results = []
+        if rank == 0:
+            for i in range(12):
+                node.submit(make_req(i, n_prompt=40, max_tokens=4))
+        for _ in range(300):
+            node.step()
+            if rank == 0:
+                results.extend(node.drain_completions())
+                done = torch.tensor([1 if len(results) >= 12 else 0])
+            else:
+                done = torch.tensor([0])
+            dist.broadcast(done, src=0, group=ctrl_group)
+            if done.item():
+                break
+        if rank == 0:
+            # DP spread: all four decode ranks served work
+            served = torch.zeros(world_size)
+        node.shutdown()
+        if rank == 0:
+            with open(out_file, "w") as f:
+                json.dump({"n": len(results),
+                           "errors": [c.error for c in results
+                                      if c.error]}, f)
+        dist.barrier(group=ctrl_group)
+    finally:
+        dist.destroy_process_group()
+
+
+@pytest.mark.timeout(240)
+class TestMonoWorld4:
+    def test_mono_dp4_pipelined(self, tmp_path):
+        init_file = str(tmp_path / "pgm4_init")
+        out_file = str(tmp_path / "outm4.json")
+        mp.start_processes(_mono4_worker, args=(4, init_file, out_file),
+                           nprocs=4, join=True, start_method="spawn")
+        with open(out_file) as f:
+            res = json.load(f)
+        assert res["n"] == 12 and not res["errors"]

@@ -35,11 +35,13 @@ hipError_t lds_gather_blocks(const void*, void*, const int32_t*, int, int,
                              int64_t, int64_t, int, hipStream_t);
 hipError_t lds_paged_attention(const void*, const void*, const void*,
                                const int32_t*, const int32_t*, void*, int, int,
-                               int, int, int, int, int, float, hipStream_t);
+                               int, int, int, int, int, int, float,
+                               hipStream_t);
 hipError_t lds_paged_attention_split(const void*, const void*, const void*,
                                      const int32_t*, const int32_t*, void*,
                                      float*, float*, int, int, int, int, int,
-                                     int, int, int, int, float, hipStream_t);
+                                     int, int, int, int, int, float,
+                                     hipStream_t);
 hipError_t lds_flash_prefill(const void*, const void*, const void*,
                              const int32_t*, const int32_t*, const int32_t*,
                              void*, int, int, int, int, int, int, int, float,
@@ -239,7 +241,8 @@ void copy_blocks_peer(uintptr_t src_pool_ptr, torch::Tensor dst_pool,
 torch::Tensor paged_attention(torch::Tensor q, torch::Tensor k_cache,
                               torch::Tensor v_cache,
                               torch::Tensor block_tables,
-                              torch::Tensor seq_lens, double scale) {
+                              torch::Tensor seq_lens, double scale,
+                              int64_t chunk) {
   CHECK_GPU(q); CHECK_GPU(k_cache); CHECK_GPU(v_cache);
   CHECK_GPU(block_tables); CHECK_GPU(seq_lens);
   int B = (int)q.size(0), qh = (int)q.size(1), d = (int)q.size(2);
@@ -249,7 +252,7 @@ torch::Tensor paged_attention(torch::Tensor q, torch::Tensor k_cache,
       q.data_ptr(), k_cache.data_ptr(), v_cache.data_ptr(),
       block_tables.data_ptr<int32_t>(), seq_lens.data_ptr<int32_t>(),
       out.data_ptr(), B, qh, kvh, bs, d, (int)block_tables.size(1),
-      kv_fp8_flag(k_cache), (float)scale, cur_stream()));
+      kv_fp8_flag(k_cache), (int)chunk, (float)scale, cur_stream()));
   return out;
 }
 
@@ -257,7 +260,8 @@ torch::Tensor paged_attention_split(torch::Tensor q, torch::Tensor k_cache,
                                     torch::Tensor v_cache,
                                     torch::Tensor block_tables,
                                     torch::Tensor seq_lens, int64_t n_parts,
-                                    int64_t part_tokens, double scale) {
+                                    int64_t part_tokens, double scale,
+                                    int64_t chunk) {
   CHECK_GPU(q); CHECK_GPU(k_cache); CHECK_GPU(v_cache);
   CHECK_GPU(block_tables); CHECK_GPU(seq_lens);
   int B = (int)q.size(0), qh = (int)q.size(1), d = (int)q.size(2);
@@ -272,7 +276,7 @@ torch::Tensor paged_attention_split(torch::Tensor q, torch::Tensor k_cache,
       block_tables.data_ptr<int32_t>(), seq_lens.data_ptr<int32_t>(),
       out.data_ptr(), part_o.data_ptr<float>(), part_ml.data_ptr<float>(), B,
       qh, kvh, bs, d, (int)block_tables.size(1), (int)n_parts,
-      (int)part_tokens, kv_fp8_flag(k_cache), (float)scale, cur_stream()));
+      (int)part_tokens, kv_fp8_flag(k_cache), (int)chunk, (float)scale, cur_stream()));
   return out;
 }
 
@@ -330,9 +334,15 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("ipc_close", &ipc_close, "unmap a peer pool");
   m.def("copy_blocks_peer", &copy_blocks_peer,
         "one-sided xGMI pull of KV blocks from a mapped peer pool");
-  m.def("paged_attention", &paged_attention, "GQA decode attention over paged KV");
+  m.def("paged_attention", &paged_attention,
+        "GQA decode attention over paged KV", py::arg("q"),
+        py::arg("k_cache"), py::arg("v_cache"), py::arg("block_tables"),
+        py::arg("seq_lens"), py::arg("scale"), py::arg("chunk") = 256);
   m.def("paged_attention_split", &paged_attention_split,
-        "flash-decoding GQA attention with sequence partitioning");
+        "flash-decoding GQA attention with sequence partitioning",
+        py::arg("q"), py::arg("k_cache"), py::arg("v_cache"),
+        py::arg("block_tables"), py::arg("seq_lens"), py::arg("n_parts"),
+        py::arg("part_tokens"), py::arg("scale"), py::arg("chunk") = 256);
   m.def("flash_prefill", &flash_prefill,
         "fused MFMA causal varlen prefill attention over paged KV");
 }

@@ -25,11 +25,11 @@
 namespace {
 
 constexpr int D = 128;        // head_dim (Llama-3 family)
-constexpr int CHUNK = 256;    // tokens per online-softmax chunk
+constexpr int DEF_CHUNK = 256;  // default tokens per online-softmax chunk
 constexpr int NW = 4;         // waves per workgroup
 constexpr float NEG = -1e30f;
 
-template <int QPG, bool SPLIT, typename CT>
+template <int QPG, bool SPLIT, typename CT, int CHUNK = DEF_CHUNK>
 __global__ __launch_bounds__(NW * WAVE) void paged_attention_kernel(
     const short* __restrict__ q,        // [B, QH, D]
     const CT* __restrict__ k_cache,     // [NB, KVH, BS, D] bf16|fp8
@@ -90,9 +90,10 @@ __global__ __launch_bounds__(NW * WAVE) void paged_attention_kernel(
 
   for (int chunk0 = t_begin; chunk0 < t_end; chunk0 += CHUNK) {
     const int n_t = min(CHUNK, t_end - chunk0);
-    // ---- Phase A: logits[h][t_local] ----
-    {
-      const int t_local = wave * WAVE + lane;
+    // ---- Phase A: logits[h][t_local] (CHUNK may exceed one row/lane:
+    // each pass of the tb loop covers NW*WAVE tokens) ----
+    for (int tb = 0; tb < CHUNK; tb += NW * WAVE) {
+      const int t_local = tb + wave * WAVE + lane;
       float dot[QPG];
 #pragma unroll
       for (int h = 0; h < QPG; ++h) dot[h] = 0.f;
@@ -117,7 +118,7 @@ __global__ __launch_bounds__(NW * WAVE) void paged_attention_kernel(
       }
 #pragma unroll
       for (int h = 0; h < QPG; ++h)
-        logits[h][wave * WAVE + lane] = (t_local < n_t) ? dot[h] : NEG;
+        logits[h][t_local] = (t_local < n_t) ? dot[h] : NEG;
     }
     __syncthreads();
 
@@ -156,8 +157,8 @@ __global__ __launch_bounds__(NW * WAVE) void paged_attention_kernel(
       o_acc[h][1] *= a;
     }
     {
-      const int t_base = wave * WAVE;
-      const int t_cnt = min(WAVE, n_t - t_base);
+      const int t_base = wave * (CHUNK / NW);
+      const int t_cnt = min(CHUNK / NW, n_t - t_base);
       // batch the V row loads 8 deep so they overlap (each row is one
       // coalesced 256-B wave read; a load-use loop serializes on latency)
       int i = 0;
@@ -287,18 +288,28 @@ hipError_t lds_paged_attention(const void* q, const void* k_cache,
                                const void* v_cache, const int32_t* block_tables,
                                const int32_t* seq_lens, void* out, int n_seqs,
                                int n_q_heads, int kvh, int bs, int head_dim,
-                               int max_blocks, int kv_fp8, float scale,
-                               hipStream_t stream) {
+                               int max_blocks, int kv_fp8, int chunk,
+                               float scale, hipStream_t stream) {
   if (n_seqs == 0) return hipSuccess;
   if (head_dim != D) return hipErrorInvalidValue;
+  if (chunk != 256 && chunk != 512) return hipErrorInvalidValue;
   const int qpg = n_q_heads / kvh;
   dim3 grid(n_seqs, kvh), block(NW * WAVE);
 #define LAUNCH_CT(QPG, CT)                                                    \
-  hipLaunchKernelGGL((paged_attention_kernel<QPG, false, CT>), grid, block,   \
-                     0, stream, (const short*)q, (const CT*)k_cache,          \
-                     (const CT*)v_cache, block_tables, seq_lens,              \
-                     (short*)out, nullptr, nullptr, kvh, bs, max_blocks, 0,   \
-                     scale)
+  do {                                                                        \
+    if (chunk == 512)                                                         \
+      hipLaunchKernelGGL((paged_attention_kernel<QPG, false, CT, 512>),       \
+                         grid, block, 0, stream, (const short*)q,             \
+                         (const CT*)k_cache, (const CT*)v_cache,              \
+                         block_tables, seq_lens, (short*)out, nullptr,        \
+                         nullptr, kvh, bs, max_blocks, 0, scale);             \
+    else                                                                      \
+      hipLaunchKernelGGL((paged_attention_kernel<QPG, false, CT, 256>),       \
+                         grid, block, 0, stream, (const short*)q,             \
+                         (const CT*)k_cache, (const CT*)v_cache,              \
+                         block_tables, seq_lens, (short*)out, nullptr,        \
+                         nullptr, kvh, bs, max_blocks, 0, scale);             \
+  } while (0)
 #define LAUNCH(QPG)                                                           \
   do {                                                                        \
     if (kv_fp8) LAUNCH_CT(QPG, unsigned char);                                \
@@ -322,18 +333,28 @@ hipError_t lds_paged_attention_split(
     const int32_t* block_tables, const int32_t* seq_lens, void* out,
     float* part_o, float* part_ml, int n_seqs, int n_q_heads, int kvh, int bs,
     int head_dim, int max_blocks, int n_parts, int part_tokens, int kv_fp8,
-    float scale, hipStream_t stream) {
+    int chunk, float scale, hipStream_t stream) {
   if (n_seqs == 0) return hipSuccess;
   if (head_dim != D || n_parts > 64) return hipErrorInvalidValue;
+  if (chunk != 256 && chunk != 512) return hipErrorInvalidValue;
   const int qpg = n_q_heads / kvh;
   dim3 grid(n_seqs, kvh, n_parts), block(NW * WAVE);
   dim3 cgrid(n_seqs, kvh), cblock(256);
 #define LAUNCH_CT(QPG, CT)                                                    \
-  hipLaunchKernelGGL((paged_attention_kernel<QPG, true, CT>), grid, block,    \
-                     0, stream, (const short*)q, (const CT*)k_cache,          \
-                     (const CT*)v_cache, block_tables, seq_lens,              \
-                     (short*)out, part_o, part_ml, kvh, bs, max_blocks,       \
-                     part_tokens, scale)
+  do {                                                                        \
+    if (chunk == 512)                                                         \
+      hipLaunchKernelGGL((paged_attention_kernel<QPG, true, CT, 512>), grid,  \
+                         block, 0, stream, (const short*)q,                   \
+                         (const CT*)k_cache, (const CT*)v_cache,              \
+                         block_tables, seq_lens, (short*)out, part_o,         \
+                         part_ml, kvh, bs, max_blocks, part_tokens, scale);   \
+    else                                                                      \
+      hipLaunchKernelGGL((paged_attention_kernel<QPG, true, CT, 256>), grid,  \
+                         block, 0, stream, (const short*)q,                   \
+                         (const CT*)k_cache, (const CT*)v_cache,              \
+                         block_tables, seq_lens, (short*)out, part_o,         \
+                         part_ml, kvh, bs, max_blocks, part_tokens, scale);   \
+  } while (0)
 #define LAUNCH(QPG)                                                           \
   do {                                                                        \
     if (kv_fp8) LAUNCH_CT(QPG, unsigned char);                                \

@@ -65,6 +65,13 @@ def reshape_and_cache(k_new: torch.Tensor, v_new: torch.Tensor,
     ref.reshape_and_cache(k_new, v_new, k_cache, v_cache, slots)
 
 
+import os as _os
+
+# online-softmax chunk of the decode kernel: 512 halves barrier count
+# (bisect knob; 256 is the measured default)
+_ATTN_CHUNK = int(_os.environ.get("LLMD_ATTN_CHUNK", "256"))
+
+
 def paged_attention(q: torch.Tensor, k_cache: torch.Tensor,
                     v_cache: torch.Tensor, block_tables: torch.Tensor,
                     seq_lens: torch.Tensor, scale: float,
@@ -81,13 +88,16 @@ def paged_attention(q: torch.Tensor, k_cache: torch.Tensor,
             target_np = min(64, max(1, 1024 // max(1, n_wgs)))
             # partition size: ceil-divide then round up to the kernel's
             # 256-token online-softmax chunk
-            part = max(256, ((max_seq_len + target_np - 1) // target_np
-                             + 255) // 256 * 256)
+            c = _ATTN_CHUNK
+            part = max(c, ((max_seq_len + target_np - 1) // target_np
+                           + c - 1) // c * c)
             np_ = (max_seq_len + part - 1) // part
             if np_ > 1:
                 return _ext.paged_attention_split(q, k_cache, v_cache, bt,
-                                                  sl, np_, part, scale)
-        return _ext.paged_attention(q, k_cache, v_cache, bt, sl, scale)
+                                                  sl, np_, part, scale,
+                                                  _ATTN_CHUNK)
+        return _ext.paged_attention(q, k_cache, v_cache, bt, sl, scale,
+                                    _ATTN_CHUNK)
     return ref.paged_attention(q, k_cache, v_cache, block_tables, seq_lens,
                                scale)
 

@@ -81,3 +81,30 @@ if __name__ == "__main__":
     bench(8, 8192, np_=16, part=512)
     bench(128, 1536)
 
+
+
+def bench_chunk(B, ctx, chunk, iters=50, qpg=4):
+    q, kc, vc, bt, sl = make(B, ctx, qpg)
+    scale = D ** -0.5
+    ref = ext.paged_attention(q, kc, vc, bt, sl, scale, 256)
+    got = ext.paged_attention(q, kc, vc, bt, sl, scale, chunk)
+    diff = (ref.float() - got.float()).abs().max().item()
+    for _ in range(10):
+        ext.paged_attention(q, kc, vc, bt, sl, scale, chunk)
+    torch.cuda.synchronize()
+    t0 = time.perf_counter()
+    for _ in range(iters):
+        ext.paged_attention(q, kc, vc, bt, sl, scale, chunk)
+    torch.cuda.synchronize()
+    us = (time.perf_counter() - t0) / iters * 1e6
+    kv_gb = B * ctx * KVH * D * 2 * 2 / 1e9
+    print(f"B={B:4d} ctx={ctx:5d} chunk={chunk}: {us:8.1f} us "
+          f"({kv_gb / (us / 1e6) / 1e3:6.2f} TB/s) max|d|={diff:.4f}")
+
+
+if __name__ == "__main__" and "--chunks" in sys.argv:
+    for B in (64, 128, 256):
+        for chunk in (256, 512):
+            bench_chunk(B, 1152, chunk)
+    bench_chunk(128, 2048, 256)
+    bench_chunk(128, 2048, 512)

@@ -72,7 +72,7 @@ def bench(B, ctx, np_=None, part=512, iters=50, kernel="v1", qpg=4):
           f"({kv_gb / (us / 1e6) / 1e3:6.2f} TB/s effective)")
 
 
-if __name__ == "__main__":
+if __name__ == "__main__" and "--chunks" not in sys.argv:
     for B in (64, 128, 256):
         bench(B, 1152)
         for np_ in (2, 4):

@@ -187,6 +187,10 @@ def build_app(service: NodeService,
     async def responses(request: Request):
         return await _handle(request, "/v1/responses")
 
+    @app.post("/v1/conversations")
+    async def conversations(request: Request):
+        return await _handle(request, "/v1/conversations")
+
     @app.post("/v1/embeddings")
     async def embeddings(request: Request):
         return await _handle(request, "/v1/embeddings")

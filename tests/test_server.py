@@ -319,3 +319,12 @@ class TestTokenizerSurface:
         ctx = SchedulingContext(request=req)
         p.produce(ctx, [])
         assert req.prompt_tokens and p.tokenizer.errors == 1
+
+
+class TestConversationsRoute:
+    def test_conversations_path(self, client):
+        r = client.post("/v1/conversations", json={
+            "model": "tiny-llama",
+            "messages": [{"role": "user", "content": "hi there friend"}],
+            "max_tokens": 3})
+        assert r.status_code == 200

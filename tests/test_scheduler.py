@@ -310,3 +310,101 @@ class TestOtlpExport:
                         span_id="b" * 16))
         ex.flush()
         assert ex.dropped == 1 and ex.sent == 0
+
+
+class TestContextLengthAwareScorer:
+    """scorer/contextlengthaware parity: in-range (0.3,1.0], out [0,0.3),
+    unlabeled neutral 0.5."""
+
+    def _eps(self):
+        from llm_d_inference_scheduler_amd.datalayer.datastore import \
+            make_endpoint
+        short = make_endpoint("short", 0, labels={
+            "llm-d.ai/context-length-range": "0-2048"})
+        long_ = make_endpoint("long", 1, labels={
+            "llm-d.ai/context-length-range": "2048-32768"})
+        plain = make_endpoint("plain", 2)
+        bad = make_endpoint("bad", 3, labels={
+            "llm-d.ai/context-length-range": "oops"})
+        return [short, long_, plain, bad]
+
+    def _score(self, n_tokens):
+        from llm_d_inference_scheduler_amd.plugins.scorers import \
+            ContextLengthAwareScorer
+        from llm_d_inference_scheduler_amd.scheduling.types import (
+            LLMRequest, SchedulingContext)
+        req = LLMRequest(request_id="r", model="m",
+                         prompt_tokens=list(range(n_tokens)))
+        return ContextLengthAwareScorer("cl").score(
+            SchedulingContext(request=req), self._eps())
+
+    def test_short_prompt_prefers_short_range(self):
+        s = self._score(256)
+        assert s["short"] > 0.3          # in range
+        assert s["long"] < 0.3           # out of range
+        assert s["plain"] == 0.5         # unlabeled neutral
+        assert s["bad"] == 0.5           # malformed label neutral
+
+    def test_long_prompt_prefers_long_range(self):
+        s = self._score(8000)
+        assert s["long"] > 0.3 and s["short"] < 0.3
+
+    def test_tighter_fit_scores_higher(self):
+        # near the top of the short range beats the bottom of it
+        hi = self._score(2000)["short"]
+        lo = self._score(64)["short"]
+        assert hi > lo > 0.3
+
+
+class TestLatencySLOAdmitter:
+    def _ctx(self, priority=0, headroom=None):
+        from llm_d_inference_scheduler_amd.datalayer.attributes import (
+            LATENCY_PREDICTION_INFO, LatencyPredictionInfo)
+        from llm_d_inference_scheduler_amd.scheduling.types import (
+            LLMRequest, SchedulingContext)
+        req = LLMRequest(request_id="r", model="m", priority=priority)
+        ctx = SchedulingContext(request=req)
+        if headroom is not None:
+            info = LatencyPredictionInfo()
+            info.ttft_headroom_ms.update(headroom)
+            ctx.attributes[LATENCY_PREDICTION_INFO] = info
+        return ctx
+
+    def _eps(self):
+        from llm_d_inference_scheduler_amd.datalayer.datastore import \
+            make_endpoint
+        return [make_endpoint("a", 0), make_endpoint("b", 1)]
+
+    def test_critical_always_admitted(self):
+        from llm_d_inference_scheduler_amd.plugins.admitters import \
+            LatencySLOAdmitter
+        adm = LatencySLOAdmitter("slo")
+        ok, _ = adm.admit(self._ctx(priority=5,
+                                    headroom={"a": -100, "b": -100}),
+                          self._eps())
+        assert ok
+
+    def test_sheddable_rejected_with_no_headroom(self):
+        from llm_d_inference_scheduler_amd.plugins.admitters import \
+            LatencySLOAdmitter
+        adm = LatencySLOAdmitter("slo")
+        ok, reason = adm.admit(self._ctx(priority=-1,
+                                         headroom={"a": -5, "b": -1}),
+                               self._eps())
+        assert not ok and "SLO" in reason
+
+    def test_admitted_when_any_endpoint_has_headroom(self):
+        from llm_d_inference_scheduler_amd.plugins.admitters import \
+            LatencySLOAdmitter
+        adm = LatencySLOAdmitter("slo")
+        ok, _ = adm.admit(self._ctx(priority=-1,
+                                    headroom={"a": -5, "b": 40}),
+                          self._eps())
+        assert ok
+
+    def test_fail_open_without_predictions(self):
+        from llm_d_inference_scheduler_amd.plugins.admitters import \
+            LatencySLOAdmitter
+        adm = LatencySLOAdmitter("slo")
+        ok, _ = adm.admit(self._ctx(priority=-1), self._eps())
+        assert ok

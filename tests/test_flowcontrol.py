@@ -422,3 +422,55 @@ class TestFlowLifecycle:
         fc.tick()
         reg.gc_flows()
         assert not reg.flows                   # all lifecycle records GC'd
+
+
+class TestConcurrentActors:
+    """Threaded-actor stress: concurrent enqueue_and_wait callers against
+    live shard threads (the reference's -race discipline for the
+    exactly-once finalization handoff, controller/doc.go)."""
+
+    @pytest.mark.timeout(120)
+    def test_concurrent_enqueue_exactly_once(self):
+        reg = FlowRegistry(bands=[BandConfig(10), BandConfig(0)],
+                           num_shards=3, flow_idle_ttl_s=0.1)
+        dispatched = []
+        lock = threading.Lock()
+
+        def dispatch(item):
+            with lock:
+                dispatched.append(item.request.request_id)
+            return True
+
+        fc = FlowController(reg, dispatch)
+        fc.start()
+        try:
+            outcomes = {}
+            olock = threading.Lock()
+
+            def caller(i):
+                it = mk_req(f"c{i}", flow=f"f{i % 7}",
+                            priority=10 if i % 2 else 0, size=1 + i % 9)
+                out = fc.enqueue_and_wait(it, timeout=30.0)
+                with olock:
+                    outcomes[f"c{i}"] = out
+            threads = [threading.Thread(target=caller, args=(i,))
+                       for i in range(120)]
+            for t in threads:
+                t.start()
+            # scale shards while the callers are in flight
+            fc.set_shard_count(1)
+            fc.set_shard_count(4)
+            for t in threads:
+                t.join(60.0)
+            assert len(outcomes) == 120
+            assert all(o == QueueOutcome.DISPATCHED
+                       for o in outcomes.values()), outcomes
+            # exactly-once dispatch: no duplicates
+            assert sorted(dispatched) == sorted(outcomes)
+            assert reg.stats.dispatched == 120
+        finally:
+            fc.stop()
+        # leases all released -> lifecycle records collectable
+        time.sleep(0.12)
+        reg.gc_flows()
+        assert not reg.flows

@@ -269,3 +269,46 @@ class TestExtProcTrailers:
         r = st.recv()
         assert r.WhichOneof("response") == "response_trailers"
         st.close()
+
+
+class TestExtProcSSEResponses:
+    def test_sse_chunks_rewritten_per_chunk(self, node, server):
+        """Streaming upstream responses: each SSE chunk's served model
+        name is rewritten back independently (server.go rewriteModelName
+        applies per streamed chunk)."""
+        from llm_d_inference_scheduler_amd.api.modelrewrite import (
+            InferenceModelRewrite, RewriteRule, RewriteTarget)
+        node.datastore.put_model_rewrite(InferenceModelRewrite(
+            name="rw", rules=[RewriteRule(
+                model="tiny-llama",
+                targets=[RewriteTarget("tiny-llama-x", weight=1)])]))
+        st = Stream(server.port)
+        st.send(req_headers({":path": "/v1/chat/completions",
+                             "content-type": "application/json",
+                             "x-request-id": "req-sse"}))
+        st.send(req_body(json.dumps({
+            "model": "tiny-llama", "stream": True,
+            "messages": [{"role": "user", "content": "hi"}],
+            "max_tokens": 4}).encode(), end_of_stream=True))
+        st.recv()               # headers response
+        st.recv()               # mutated body (rewrite)
+        st.send(resp_headers())
+        st.recv()
+        chunk1 = ('data: ' + json.dumps(
+            {"model": "tiny-llama-x",
+             "choices": [{"delta": {"content": "a"}}]}) + "\n\n").encode()
+        st.send(resp_body(chunk1))
+        r = st.recv()
+        cr = r.response_body.response
+        assert cr.status == pb.CONTINUE_AND_REPLACE
+        out = cr.body_mutation.body.decode()
+        assert '"model": "tiny-llama"' in out and "tiny-llama-x" not in out
+        # final chunk with usage closes the stream accounting
+        final = ("data: " + json.dumps(
+            {"model": "tiny-llama-x",
+             "usage": {"prompt_tokens": 1, "completion_tokens": 4}})
+            + "\n\ndata: [DONE]\n\n").encode()
+        st.send(resp_body(final, end_of_stream=True))
+        r = st.recv()
+        assert r.WhichOneof("response") == "response_body"
+        st.close()

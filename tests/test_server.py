@@ -328,3 +328,35 @@ class TestConversationsRoute:
             "messages": [{"role": "user", "content": "hi there friend"}],
             "max_tokens": 3})
         assert r.status_code == 200
+
+
+class TestServeCLI:
+    """The `python -m ...server` entry (cmd/epp/main.go analog)."""
+
+    def test_build_node_from_flags(self):
+        from llm_d_inference_scheduler_amd.server.__main__ import (
+            build_node, parse_args)
+        args = parse_args(["--model", "tiny-llama", "--device", "cpu",
+                           "--flow-control", "--kv-dtype", "bf16"])
+        node = build_node(args)
+        assert node.is_router and node.flow is not None
+        node.shutdown()
+
+    def test_config_text_flag(self):
+        from llm_d_inference_scheduler_amd.server.__main__ import (
+            build_node, parse_args)
+        yaml_text = """
+plugins:
+  - type: queue-scorer
+  - type: max-score-picker
+schedulingProfiles:
+  - name: default
+    plugins:
+      - {pluginRef: queue-scorer, weight: 1}
+      - {pluginRef: max-score-picker}
+"""
+        args = parse_args(["--model", "tiny-llama", "--device", "cpu",
+                           "--config-text", yaml_text])
+        node = build_node(args)
+        assert "default" in node.loaded.scheduler_config.profiles
+        node.shutdown()

@@ -272,3 +272,62 @@ class TestStopTokens:
         assert fin.completion_tokens == 3
         assert fin.all_tokens == ref[:3]
         assert not w2.has_work and w2.mgr.usage == pytest.approx(0.0)
+
+
+class TestPriorityPreemptsBeforeExhaustion:
+    def test_critical_waiter_preempts_running_low_priority(self):
+        """A higher-priority arrival whose prefill cannot allocate
+        preempts the lowest-priority running request immediately instead
+        of waiting for full KV exhaustion (round-2 refinement)."""
+        w = make_worker(kv_blocks=16)   # 256 token slots
+        # low-priority request grabs most of the pool, with a generation
+        # long enough that it cannot finish within the probe window
+        w.add_request(EngineRequest("low", list(range(150)), max_tokens=90,
+                                    priority=-1))
+        for _ in range(20):
+            w.step()
+            if w.running:
+                break
+        assert w.running and w.running[0].request_id == "low"
+        # collect pending so the victim is preemptible
+        for _ in range(3):
+            w.step()
+        # critical arrival needs more than the free space
+        w.add_request(EngineRequest("crit", list(range(200, 300)),
+                                    max_tokens=4, priority=10))
+        preempted = False
+        outs = []
+        for _ in range(60):
+            outs.extend(w.step())
+            if any(r.request_id == "crit" for r in w.running):
+                preempted = True
+                break
+        assert preempted, "critical request never started decoding"
+        # the low-priority request was requeued (recompute), not lost
+        ids_active = {r.request_id for r in w.running} | \
+                     {r.request_id for r in w.waiting}
+        assert "low" in ids_active
+        # run everything to completion: both finish
+        for _ in range(300):
+            outs.extend(w.step())
+            if not w.has_work:
+                break
+        fin = {o.request_id for o in outs if o.finished and not o.error}
+        assert {"low", "crit"} <= fin
+
+    def test_no_preemption_for_equal_priority(self):
+        w = make_worker(kv_blocks=16)
+        w.add_request(EngineRequest("a", list(range(150)), max_tokens=90,
+                                    priority=0))
+        for _ in range(20):
+            w.step()
+            if w.running:
+                break
+        for _ in range(3):
+            w.step()
+        w.add_request(EngineRequest("b", list(range(200, 300)),
+                                    max_tokens=4, priority=0))
+        for _ in range(10):
+            w.step()
+        # same priority never displaces: "a" keeps running
+        assert any(r.request_id == "a" for r in w.running)

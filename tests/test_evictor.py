@@ -92,3 +92,45 @@ class TestNodeInflightEviction:
         assert node.evictor.stats[1] == 0
         assert "crit" not in comps            # critical keeps running
         node.shutdown()
+
+
+class TestExtProcEvictionWiring:
+    def test_inflight_eviction_reaches_extproc_stream(self):
+        """node._maybe_evict_inflight notifies the attached ext-proc
+        server so the open stream gets its 429 (server.go:262-284)."""
+        import torch
+        from llm_d_inference_scheduler_amd.flowcontrol.evictor import \
+            EvictionItem
+        from llm_d_inference_scheduler_amd.models.configs import TINY_LLAMA
+        from llm_d_inference_scheduler_amd.node import NodeConfig, NodeRunner
+        node = NodeRunner(NodeConfig(model=TINY_LLAMA, device="cpu",
+                                     dtype=torch.float32, kv_blocks=64,
+                                     flow_control=True))
+        try:
+            evicted = []
+
+            class FakeExtProc:
+                def evict(self, rid, reason="evicted"):
+                    evicted.append(rid)
+                    return True
+            node.extproc = FakeExtProc()
+            node.evictor.track(EvictionItem(request_id="victim",
+                                            priority=-1, target="gpu0"))
+            # force the saturation conditions
+            node.detector.is_saturated = lambda eps: True
+            node.flow.submitedness = None
+            # one queued item so the evictor path arms
+            from llm_d_inference_scheduler_amd.flowcontrol.types import \
+                FlowControlRequest
+            from llm_d_inference_scheduler_amd.scheduling.types import \
+                LLMRequest
+            # block dispatch so the item stays queued
+            for s in node.flow.shards:
+                s.dispatch_fn = lambda item: False
+            node.flow.submit(FlowControlRequest(
+                request=LLMRequest(request_id="q1", model="m", prompt="x"),
+                flow_key="f", byte_size=1))
+            node._maybe_evict_inflight()
+            assert evicted == ["victim"]
+        finally:
+            node.shutdown()

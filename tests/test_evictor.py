@@ -118,7 +118,7 @@ class TestExtProcEvictionWiring:
                                             priority=-1, target="gpu0"))
             # force the saturation conditions
             node.detector.is_saturated = lambda eps: True
-            node.flow.submitedness = None
+            
             # one queued item so the evictor path arms
             from llm_d_inference_scheduler_amd.flowcontrol.types import \
                 FlowControlRequest

@@ -190,13 +190,15 @@ class TestEngineLifecycleProperty:
               suppress_health_check=[HealthCheck.too_slow])
     @given(st.lists(st.tuples(st.integers(1, 60),      # prompt len
                               st.integers(1, 6),       # max_tokens
-                              st.booleans()),          # abort it mid-flight?
+                              st.booleans(),           # abort it mid-flight?
+                              st.integers(-1, 10)),    # priority
                     min_size=1, max_size=5),
            st.integers(6, 24))                          # pool blocks
     def test_requests_always_terminate(self, reqs, blocks):
         """Every submitted request either completes or is aborted — no
         stranded sequences, and the pool leaks no blocks (vLLM-style
-        invariant over random shapes incl. pool churn/preemption)."""
+        invariant over random shapes incl. pool churn, exhaustion
+        preemption AND priority-driven early preemption)."""
         import torch
         from llm_d_inference_scheduler_amd.engine import (EngineRequest,
                                                           EngineWorker)
@@ -204,10 +206,10 @@ class TestEngineLifecycleProperty:
         w = EngineWorker(TINY_LLAMA, "cpu", kv_blocks=blocks,
                          dtype=torch.float32)
         done = set()
-        for i, (plen, mx, abort) in enumerate(reqs):
+        for i, (plen, mx, abort, prio) in enumerate(reqs):
             w.add_request(EngineRequest(f"r{i}", list(range(plen)),
-                                        max_tokens=mx))
-        aborted = {f"r{i}" for i, (_, _, a) in enumerate(reqs) if a}
+                                        max_tokens=mx, priority=prio))
+        aborted = {f"r{i}" for i, (_, _, a, _) in enumerate(reqs) if a}
         stepped = 0
         while w.has_work and stepped < 400:
             if stepped == 2 and aborted:

@@ -65,6 +65,11 @@ def _headers_into(req: LLMRequest, headers: Dict[str, str]) -> None:
     req.headers.update(headers)
     req.fairness_id = headers.get(FAIRNESS_ID_HEADER, "")
     req.objective_name = headers.get(OBJECTIVE_HEADER, "")
+    # explicit per-request rewrite wins over InferenceModelRewrite rules
+    # (request.go:56 ModelNameRewriteKey -> TargetModelName)
+    forced = headers.get(MODEL_REWRITE_HEADER, "")
+    if forced:
+        req.target_model = forced
 
 
 def _extract_mm_items(messages: List[Dict[str, Any]]) -> List[MultiModalItem]:

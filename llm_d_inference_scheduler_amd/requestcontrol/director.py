@@ -170,6 +170,11 @@ class Director:
                                    ctx=ctx, epp_latency_ms=latency_ms)
 
     def _mutate_model(self, req: LLMRequest) -> None:
+        # a header-forced target (x-gateway-model-name-rewrite) wins over
+        # InferenceModelRewrite rules (director.go mutateModelIfNeeded
+        # skips when TargetModelName was set from the header)
+        if req.target_model != req.model:
+            return
         for rw in self.datastore.model_rewrites():
             rule = rw.match(req.model)
             if rule is None:

@@ -154,3 +154,41 @@ class TestVllmGrpcParser:
             b"\x00\x00\x00\x00\x03\xff\xff\xff", {},
             "/vllm.VllmEngine/Generate")
         assert res.error is not None
+
+
+class TestModelRewriteHeader:
+    def test_header_forces_target_model(self):
+        from llm_d_inference_scheduler_amd.handlers.parsers import (
+            MODEL_REWRITE_HEADER, OpenAIParser)
+        body = json.dumps({"model": "m", "prompt": "x"}).encode()
+        r = OpenAIParser("p").parse_request(
+            body, {MODEL_REWRITE_HEADER: "m-forced"}, "/v1/completions")
+        assert r.request.target_model == "m-forced"
+        assert r.request.model == "m"
+
+    def test_header_wins_over_crd_rewrite(self):
+        import torch
+        from llm_d_inference_scheduler_amd.api.modelrewrite import (
+            InferenceModelRewrite, RewriteRule, RewriteTarget)
+        from llm_d_inference_scheduler_amd.handlers.parsers import \
+            MODEL_REWRITE_HEADER
+        from llm_d_inference_scheduler_amd.models.configs import TINY_LLAMA
+        from llm_d_inference_scheduler_amd.node import NodeConfig, NodeRunner
+        from llm_d_inference_scheduler_amd.scheduling.types import LLMRequest
+        node = NodeRunner(NodeConfig(model=TINY_LLAMA, device="cpu",
+                                     dtype=torch.float32, kv_blocks=64))
+        node.datastore.put_model_rewrite(InferenceModelRewrite(
+            name="rw", rules=[RewriteRule(
+                model="tiny-llama",
+                targets=[RewriteTarget("crd-target", weight=1)])]))
+        req = LLMRequest(request_id="h1", model="tiny-llama", prompt="x",
+                         headers={MODEL_REWRITE_HEADER: "hdr-target"},
+                         target_model="hdr-target")
+        d = node.director.handle_request(req)
+        assert d.request.target_model == "hdr-target"
+        node.director.handle_response_complete(d, None)
+        # without the header, the CRD rule applies
+        req2 = LLMRequest(request_id="h2", model="tiny-llama", prompt="x")
+        d2 = node.director.handle_request(req2)
+        assert d2.request.target_model == "crd-target"
+        node.shutdown()

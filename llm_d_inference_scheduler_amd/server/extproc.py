@@ -270,6 +270,7 @@ class ExtProcServer:
             yield from self._body_responses(None)
             return
         req = result.request
+        prom.L(prom.request_sizes, req.model).observe(len(body))
         req.request_id = st.request_id
         req.headers.update(st.headers)
         if FAIRNESS_HEADER in st.headers:

@@ -82,6 +82,7 @@ def build_app(service: NodeService,
         if result.skip or req is None:
             return _error(400, "unparseable", "passthrough not routable "
                           "without an upstream")
+        prom.L(prom.request_sizes, req.model).observe(len(body))
         if req.prompt_tokens is None and not req.prompt and req.messages:
             pass  # token-producer will tokenize messages
         raw_stop = (result.request.raw_body or {}).get("stop") \

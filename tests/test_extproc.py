@@ -312,3 +312,40 @@ class TestExtProcSSEResponses:
         r = st.recv()
         assert r.WhichOneof("response") == "response_body"
         st.close()
+
+
+class TestExtProcConcurrentStreams:
+    def test_parallel_streams_route_independently(self, server):
+        """Several simultaneous Process streams (the gRPC thread pool +
+        director scheduling lock): each gets its own routed response."""
+        def drive(i, results):
+            st = Stream(server.port)
+            try:
+                st.send(req_headers({":path": "/v1/completions",
+                                     "content-type": "application/json",
+                                     "x-request-id": f"par-{i}"}))
+                st.send(req_body(completion_body(
+                    prompt=f"prompt {i} " * 8), end_of_stream=True))
+                r1 = st.recv()
+                ok = r1.WhichOneof("response") == "request_headers"
+                r2 = st.recv()
+                ok = ok and r2.WhichOneof("response") == "request_body"
+                st.send(resp_headers())
+                st.recv()
+                st.send(resp_body(b'{"model":"tiny-llama","usage":'
+                                  b'{"prompt_tokens":1,'
+                                  b'"completion_tokens":1}}',
+                                  end_of_stream=True))
+                r4 = st.recv()
+                ok = ok and r4.WhichOneof("response") == "response_body"
+                results[i] = ok
+            finally:
+                st.close()
+        results = {}
+        threads = [threading.Thread(target=drive, args=(i, results))
+                   for i in range(8)]
+        for t in threads:
+            t.start()
+        for t in threads:
+            t.join(20.0)
+        assert len(results) == 8 and all(results.values()), results

@@ -36,7 +36,8 @@ def parse_args():
                    choices=["mono", "pd", "fc", "epd"],
                    help="mono=prefix-aware DP decode; pd=P/D disagg; "
                         "fc=flow control at overload; epd=E/P/D multimodal")
-    p.add_argument("--model", default="llama-3-8b")
+    p.add_argument("--model", default="llama-3-8b",
+                   help="llama-3-8b | qwen3-32b (the reference regression model)")
     # the reference's regression harness shape: input 1024 / output 1024
     # (config/manifests/regression-testing/single-workload-regression.yaml)
     p.add_argument("--prompt-len", type=int, default=1024)
@@ -204,7 +205,7 @@ def main():
         ctrl_group = None
 
     from llm_d_inference_scheduler_amd.models.configs import (
-        LLAMA_3_8B, LLAVA_1_5_7B_TEXT, TINY_LLAMA, TINY_LLAVA)
+        LLAMA_3_8B, LLAVA_1_5_7B_TEXT, QWEN3_32B, TINY_LLAMA, TINY_LLAVA)
     from llm_d_inference_scheduler_amd.node import NodeConfig, NodeRunner
     from llm_d_inference_scheduler_amd.flowcontrol import BandConfig
     from llm_d_inference_scheduler_amd.api.objectives import InferenceObjective
@@ -214,11 +215,13 @@ def main():
         # world * (weights + KV + activations) fits
         args.kv_gb = min(args.kv_gb, max(16.0, 288.0 / world - 48.0))
 
-    model_cfg = LLAMA_3_8B if args.model == "llama-3-8b" else TINY_LLAMA
+    model_cfg = {"llama-3-8b": LLAMA_3_8B,
+                 "qwen3-32b": QWEN3_32B}.get(args.model, TINY_LLAMA)
     if args.mode == "epd":
         model_cfg = LLAVA_1_5_7B_TEXT  # BASELINE config 5 multimodal model
         args.model = model_cfg.name
-    if not use_gpu and args.model in ("llama-3-8b", "llava-1.5-7b"):
+    if not use_gpu and args.model in ("llama-3-8b", "qwen3-32b",
+                                      "llava-1.5-7b"):
         # CPU smoke of the bench harness itself uses the tiny config; a GPU
         # run always uses the full flagship model (anything else is invalid
         # for reporting).

@@ -118,6 +118,11 @@ class LlamaRunner:
                 "wo": w(c.q_size, c.hidden_size, std=out_std),
                 "post_norm": torch.ones(c.hidden_size, dtype=dtype,
                                         device=device),
+                **({"q_norm": torch.ones(c.head_dim, dtype=dtype,
+                                         device=device),
+                    "k_norm": torch.ones(c.head_dim, dtype=dtype,
+                                         device=device)}
+                   if c.qk_norm else {}),
                 "wgate_up": w(c.hidden_size, 2 * c.intermediate_size),
                 "wdown": w(c.intermediate_size, c.hidden_size, std=out_std),
             })
@@ -158,6 +163,17 @@ class LlamaRunner:
             q = q.view(T, c.num_heads, c.head_dim).contiguous()
             k = k.view(T, c.num_kv_heads, c.head_dim).contiguous()
             v = v.view(T, c.num_kv_heads, c.head_dim).contiguous()
+            if c.qk_norm:
+                # Qwen3 per-head QK-RMSNorm before RoPE: each head's
+                # D-vector normalized with a learned gain (the rmsnorm
+                # kernel treats [T*H, D] rows like any hidden dim)
+                q = ops.rmsnorm(q.view(T * c.num_heads, c.head_dim),
+                                layer["q_norm"],
+                                c.rms_eps).view(T, c.num_heads, c.head_dim)
+                k = ops.rmsnorm(k.view(T * c.num_kv_heads, c.head_dim),
+                                layer["k_norm"],
+                                c.rms_eps).view(T, c.num_kv_heads,
+                                                c.head_dim)
             ops.rope(q, k, self.cos_sin, pos32)
             k_cache = kv_pool[li, 0]
             v_cache = kv_pool[li, 1]

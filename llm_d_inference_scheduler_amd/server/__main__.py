@@ -25,7 +25,8 @@ def parse_args(argv=None):
                    help="Envoy ext-proc gRPC front door (0 = disabled; "
                         "the reference's default grpcPort is 9002)")
     p.add_argument("--model", default="llama-3-8b",
-                   choices=["llama-3-8b", "llava-1.5-7b", "tiny-llama"])
+                   choices=["llama-3-8b", "qwen3-32b", "llava-1.5-7b",
+                            "tiny-llama"])
     p.add_argument("--topology", default="mono",
                    help='"mono", "pd-combined", "pd:NpMd", "epd:1eNpMd"')
     p.add_argument("--config-file", default="",
@@ -46,9 +47,11 @@ def parse_args(argv=None):
 def build_node(args):
     import torch
 
-    from ..models.configs import LLAMA_3_8B, LLAVA_1_5_7B_TEXT, TINY_LLAMA
+    from ..models.configs import (LLAMA_3_8B, LLAVA_1_5_7B_TEXT,
+                                  QWEN3_32B, TINY_LLAMA)
     from ..node import NodeConfig, NodeRunner
-    model = {"llama-3-8b": LLAMA_3_8B, "llava-1.5-7b": LLAVA_1_5_7B_TEXT,
+    model = {"llama-3-8b": LLAMA_3_8B, "qwen3-32b": QWEN3_32B,
+             "llava-1.5-7b": LLAVA_1_5_7B_TEXT,
              "tiny-llama": TINY_LLAMA}[args.model]
     use_gpu = torch.cuda.is_available() if args.device is None else \
         str(args.device).startswith("cuda")

@@ -302,3 +302,40 @@ class TestModelLenLimits:
             # generation retained exactly (no token ever dropped)
             assert out.all_tokens[:len(gen_before)] == gen_before
             assert victim not in w.running
+
+
+class TestQwenQKNorm:
+    def test_tiny_qwen_generates_and_differs_from_no_norm(self):
+        """Qwen3-style per-head QK-RMSNorm: the qk_norm model runs end to
+        end, and toggling the flag changes the computation (the norm is
+        actually applied)."""
+        import dataclasses
+        from llm_d_inference_scheduler_amd.models.configs import TINY_QWEN
+        outs = {}
+        for qk in (True, False):
+            cfg = dataclasses.replace(TINY_QWEN, qk_norm=qk)
+            w = EngineWorker(cfg, "cpu", kv_blocks=64, dtype=torch.float32,
+                             seed=7)
+            w.add_request(EngineRequest("q", list(range(40, 90)),
+                                        max_tokens=6))
+            toks = []
+            for _ in range(40):
+                for o in w.step():
+                    toks.extend(o.new_tokens)
+                if not w.has_work:
+                    break
+            assert len(toks) == 6
+            outs[qk] = toks
+        assert outs[True] != outs[False]
+
+    def test_qwen32b_config_shape(self):
+        from llm_d_inference_scheduler_amd.models.configs import QWEN3_32B
+        assert QWEN3_32B.num_heads // QWEN3_32B.num_kv_heads == 8  # qpg
+        assert QWEN3_32B.head_dim == 128 and QWEN3_32B.qk_norm
+        # bf16 weights ~64 GB: fits one MI355X beside a large KV pool
+        approx_params = (QWEN3_32B.num_layers * (
+            QWEN3_32B.hidden_size * (QWEN3_32B.q_size + 2 * QWEN3_32B.kv_size
+                                     + QWEN3_32B.q_size)
+            + 3 * QWEN3_32B.hidden_size * QWEN3_32B.intermediate_size)
+            + 2 * QWEN3_32B.vocab_size * QWEN3_32B.hidden_size)
+        assert 30e9 < approx_params < 40e9

@@ -225,7 +225,9 @@ def main():
         # CPU smoke of the bench harness itself uses the tiny config; a GPU
         # run always uses the full flagship model (anything else is invalid
         # for reporting).
-        model_cfg = TINY_LLAVA if args.mode == "epd" else TINY_LLAMA
+        from llm_d_inference_scheduler_amd.models.configs import TINY_QWEN
+        model_cfg = (TINY_LLAVA if args.mode == "epd" else
+                     TINY_QWEN if args.model == "qwen3-32b" else TINY_LLAMA)
         args.prompt_len = min(args.prompt_len, 96)
         args.max_tokens = min(args.max_tokens, 8)
         args.concurrency = min(args.concurrency, 8)

@@ -553,3 +553,34 @@ class TestStreamOverlapGPU:
             assert set(toks) == {"a", "b", "c"}
             results[overlap] = toks
         assert results[False] == results[True]
+
+
+class TestQKNormGPU:
+    def test_qwen_qk_norm_hip_path(self):
+        """Per-head QK-RMSNorm rides the HIP rmsnorm kernel on [T*H, D]
+        rows: GPU generation matches the structure of the CPU run (same
+        count; norm toggling changes outputs)."""
+        import dataclasses
+        from llm_d_inference_scheduler_amd.engine import (EngineRequest,
+                                                          EngineWorker)
+        from llm_d_inference_scheduler_amd.models.configs import ModelConfig
+        cfg = ModelConfig(name="gpu-qk", vocab_size=2048, hidden_size=1024,
+                          intermediate_size=2048, num_layers=2, num_heads=8,
+                          num_kv_heads=2, head_dim=128, rope_theta=1e4,
+                          qk_norm=True)
+        outs = {}
+        for qk in (True, False):
+            c = dataclasses.replace(cfg, qk_norm=qk)
+            w = EngineWorker(c, "cuda:0", kv_blocks=256,
+                             dtype=torch.bfloat16, seed=13)
+            w.add_request(EngineRequest("q", list(range(300, 430)),
+                                        max_tokens=6))
+            toks = []
+            for _ in range(30):
+                for o in w.step():
+                    toks.extend(o.new_tokens)
+                if not w.has_work:
+                    break
+            assert len(toks) == 6
+            outs[qk] = toks
+        assert outs[True] != outs[False]
